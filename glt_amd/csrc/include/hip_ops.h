@@ -1,0 +1,84 @@
+/* Host-callable declarations for the HIP (gfx950) kernels.
+ *
+ * bindings.cpp is compiled by the host compiler, so everything HIP-side is
+ * exposed as plain functions + opaque handles here; implementations live in
+ * csrc/hip/*.hip.
+ */
+#pragma once
+
+#include <torch/extension.h>
+
+#include <memory>
+#include <string>
+#include <vector>
+
+namespace glt {
+
+// --- samplers (hip_sampler.hip) -------------------------------------------
+std::tuple<torch::Tensor, torch::Tensor, c10::optional<torch::Tensor>>
+hip_sample_neighbors(const torch::Tensor& indptr, const torch::Tensor& indices,
+                     const c10::optional<torch::Tensor>& edge_ids,
+                     const c10::optional<torch::Tensor>& edge_weights,
+                     const torch::Tensor& seeds, int64_t k, bool with_edge,
+                     bool weighted);
+torch::Tensor hip_lookup_degree(const torch::Tensor& indptr,
+                                const torch::Tensor& nodes);
+torch::Tensor hip_sample_negative(const torch::Tensor& indptr,
+                                  const torch::Tensor& indices,
+                                  int64_t num_cols, int64_t req_num,
+                                  int64_t trials, bool padding);
+torch::Tensor hip_random_walk(const torch::Tensor& indptr,
+                              const torch::Tensor& indices,
+                              const torch::Tensor& seeds, int64_t walk_len);
+torch::Tensor hip_cal_nbr_prob(const torch::Tensor& indptr,
+                               const torch::Tensor& indices,
+                               const torch::Tensor& last_prob,
+                               const torch::Tensor& nodes, int64_t k);
+
+// --- inducer / subgraph / stitch (hip_inducer.hip) -------------------------
+class HIPInducer;
+std::shared_ptr<HIPInducer> hip_inducer_create(int64_t reserve);
+torch::Tensor hip_inducer_init_node(HIPInducer* ind, const torch::Tensor& s);
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor>
+hip_inducer_induce_next(HIPInducer* ind, const torch::Tensor& srcs,
+                        const torch::Tensor& nbrs,
+                        const torch::Tensor& nbrs_num);
+torch::Tensor hip_inducer_lookup(HIPInducer* ind, const torch::Tensor& ids);
+int64_t hip_inducer_count(HIPInducer* ind);
+
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor,
+           c10::optional<torch::Tensor>>
+hip_node_subgraph(const torch::Tensor& indptr, const torch::Tensor& indices,
+                  const c10::optional<torch::Tensor>& edge_ids,
+                  const torch::Tensor& nodes, bool with_edge);
+
+std::tuple<torch::Tensor, torch::Tensor, c10::optional<torch::Tensor>>
+hip_stitch_sample_results(int64_t ids_count,
+                          const std::vector<torch::Tensor>& idx_list,
+                          const std::vector<torch::Tensor>& nbrs_list,
+                          const std::vector<torch::Tensor>& nbrs_num_list,
+                          const std::vector<torch::Tensor>& eids_list);
+
+// --- memory plumbing (hip_mem.hip) -----------------------------------------
+// Device-dtype alias of (pinned/registered) host memory; keeps `src` alive.
+torch::Tensor host_mapped_view(const torch::Tensor& src, int64_t device_index);
+// hipHostRegister arbitrary host range (e.g. the shm sample ring).
+void pin_host_memory(int64_t addr, int64_t bytes);
+void unpin_host_memory(int64_t addr);
+void enable_peer_access(int64_t device, int64_t peer);
+// IPC sharing of a device tensor: returns handle bytes.
+std::string ipc_share(const torch::Tensor& t);
+torch::Tensor ipc_open(const std::string& handle, int64_t device,
+                       const std::vector<int64_t>& shape,
+                       torch::ScalarType dtype);
+
+// --- unified feature store (hip_unified_tensor.hip) ------------------------
+class UnifiedFeatureStore;
+std::shared_ptr<UnifiedFeatureStore> ufs_create(int64_t device_index);
+// Append a row segment; tensor must be HIP-device or a host_mapped_view.
+void ufs_append(UnifiedFeatureStore* s, const torch::Tensor& seg);
+torch::Tensor ufs_gather(UnifiedFeatureStore* s, const torch::Tensor& rows);
+int64_t ufs_rows(UnifiedFeatureStore* s);
+int64_t ufs_dim(UnifiedFeatureStore* s);
+
+}  // namespace glt
