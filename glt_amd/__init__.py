@@ -1,0 +1,34 @@
+"""glt_amd — MI355X-native graph learning engine.
+
+A from-scratch GNN sampling-and-training framework for AMD Instinct MI355X
+(CDNA4 / gfx950) with the capabilities of alibaba/graphlearn-for-pytorch:
+GPU neighbor/negative/subgraph sampling as hand-written HIP kernels, a
+tiered feature store over HBM3E + xGMI peers + pinned-host UVA, and a
+distributed runtime (partitioned graphs, RPC sampling workers, shm sample
+channels, RCCL collectives).  See SURVEY.md for the structural map.
+"""
+__version__ = "0.1.0"
+
+import os
+
+# RCCL over xGMI: the host driver supports dmabuf IPC only.
+os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+
+from . import _C  # native core (built in-tree; fail loudly if missing)
+from . import data, loader, models, partition, sampler, utils
+from .data import Dataset, DeviceGroup, Feature, Graph, Topology
+from .loader import (LinkLoader, LinkNeighborLoader, NeighborLoader,
+                     NodeLoader, SubGraphLoader)
+from .sampler import (EdgeSamplerInput, NegativeSampling, NeighborSampler,
+                      NodeSamplerInput, RandomNegativeSampler,
+                      SamplingConfig, SamplingType)
+from .utils import seed_everything
+
+__all__ = [
+    "_C", "data", "loader", "models", "partition", "sampler", "utils",
+    "Dataset", "DeviceGroup", "Feature", "Graph", "Topology",
+    "LinkLoader", "LinkNeighborLoader", "NeighborLoader", "NodeLoader",
+    "SubGraphLoader", "EdgeSamplerInput", "NegativeSampling",
+    "NeighborSampler", "NodeSamplerInput", "RandomNegativeSampler",
+    "SamplingConfig", "SamplingType", "seed_everything",
+]

@@ -97,6 +97,9 @@ struct DeviceInducer {
   torch::Tensor lookup(const torch::Tensor& ids) {
     return hip_inducer_lookup(p.get(), ids);
   }
+  torch::Tensor insert(const torch::Tensor& ids) {
+    return hip_inducer_insert(p.get(), ids);
+  }
   int64_t count() { return hip_inducer_count(p.get()); }
 };
 
@@ -175,6 +178,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("init_node", &DeviceInducer::init_node)
       .def("induce_next", &DeviceInducer::induce_next)
       .def("lookup", &DeviceInducer::lookup)
+      .def("insert", &DeviceInducer::insert)
       .def("count", &DeviceInducer::count);
 
   py::class_<FeatureStorePy>(m, "UnifiedFeatureStore")

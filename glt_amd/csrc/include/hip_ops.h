@@ -44,6 +44,7 @@ hip_inducer_induce_next(HIPInducer* ind, const torch::Tensor& srcs,
                         const torch::Tensor& nbrs,
                         const torch::Tensor& nbrs_num);
 torch::Tensor hip_inducer_lookup(HIPInducer* ind, const torch::Tensor& ids);
+torch::Tensor hip_inducer_insert(HIPInducer* ind, const torch::Tensor& ids);
 int64_t hip_inducer_count(HIPInducer* ind);
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor,

@@ -1,0 +1,8 @@
+from .graph import Topology, Graph
+from .feature import Feature, DeviceGroup
+from .dataset import Dataset
+from .reorder import sort_by_in_degree
+from .unified_tensor import UnifiedTensor
+
+__all__ = ["Topology", "Graph", "Feature", "DeviceGroup", "Dataset",
+           "sort_by_in_degree", "UnifiedTensor"]

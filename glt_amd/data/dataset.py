@@ -1,0 +1,162 @@
+"""Dataset container (parity: reference python/data/dataset.py:30-122,
+236-394, 453-489): graph(s) + node/edge features + labels + splits, with
+builders and whole-dataset process sharing."""
+from typing import Dict, List, Optional, Tuple, Union
+
+import torch
+
+from ..typing import EdgeType, NodeType
+from .feature import Feature, DeviceGroup
+from .graph import Graph, Topology
+from .reorder import sort_by_in_degree
+
+
+class Dataset:
+    def __init__(self, graph=None, node_features=None, edge_features=None,
+                 node_labels=None, edge_dir: str = "out"):
+        self.graph: Union[Graph, Dict[EdgeType, Graph], None] = graph
+        self.node_features: Union[Feature, Dict[NodeType, Feature], None] = \
+            node_features
+        self.edge_features: Union[Feature, Dict[EdgeType, Feature], None] = \
+            edge_features
+        self.node_labels: Union[torch.Tensor,
+                                Dict[NodeType, torch.Tensor], None] = \
+            node_labels
+        self.edge_dir = edge_dir
+        self.train_idx = None
+        self.val_idx = None
+        self.test_idx = None
+
+    # -- builders -----------------------------------------------------------
+    def init_graph(self, edge_index=None, edge_ids=None, edge_weights=None,
+                   layout: str = "COO", graph_mode: str = "ZERO_COPY",
+                   device: Optional[int] = None,
+                   num_nodes: Optional[int] = None):
+        """edge_index: [2, E] (homo) or Dict[EdgeType, [2, E]] (hetero)."""
+        target = "CSC" if self.edge_dir == "in" else "CSR"
+        if isinstance(edge_index, dict):
+            self.graph = {}
+            for etype, ei in edge_index.items():
+                eid = edge_ids.get(etype) if isinstance(edge_ids, dict) else None
+                ew = (edge_weights.get(etype)
+                      if isinstance(edge_weights, dict) else None)
+                nn = num_nodes.get(etype) if isinstance(num_nodes, dict) \
+                    else num_nodes
+                topo = Topology(ei, eid, ew, input_layout=layout,
+                                layout=target, num_nodes=nn)
+                self.graph[etype] = Graph(topo, graph_mode, device)
+        elif edge_index is not None:
+            topo = Topology(edge_index, edge_ids, edge_weights,
+                            input_layout=layout, layout=target,
+                            num_nodes=num_nodes)
+            self.graph = Graph(topo, graph_mode, device)
+        return self
+
+    def init_node_features(self, node_feature_data=None,
+                           id2idx: Optional[torch.Tensor] = None,
+                           sort_func=None, split_ratio: float = 1.0,
+                           device_group_list: Optional[List[DeviceGroup]] = None,
+                           device: Optional[int] = None,
+                           with_gpu: bool = True,
+                           dtype: Optional[torch.dtype] = None):
+        def build(feat, topo, i2i):
+            if sort_func is not None and topo is not None and i2i is None:
+                feat, i2i = sort_func(feat, split_ratio, topo)
+            return Feature(feat, split_ratio, device_group_list, device,
+                           with_gpu, dtype=dtype, id2index=i2i)
+
+        if isinstance(node_feature_data, dict):
+            self.node_features = {}
+            for ntype, feat in node_feature_data.items():
+                topo = None
+                if isinstance(self.graph, dict):
+                    for (src, _, dst), g in self.graph.items():
+                        if dst == ntype:
+                            topo = g.topo
+                            break
+                i2i = id2idx.get(ntype) if isinstance(id2idx, dict) else None
+                self.node_features[ntype] = build(feat, topo, i2i)
+        elif node_feature_data is not None:
+            topo = self.graph.topo if isinstance(self.graph, Graph) else None
+            self.node_features = build(node_feature_data, topo, id2idx)
+        return self
+
+    def init_edge_features(self, edge_feature_data=None, id2idx=None,
+                           split_ratio: float = 0.0,
+                           device_group_list=None, device=None,
+                           with_gpu: bool = True,
+                           dtype: Optional[torch.dtype] = None):
+        if isinstance(edge_feature_data, dict):
+            self.edge_features = {}
+            for etype, feat in edge_feature_data.items():
+                i2i = id2idx.get(etype) if isinstance(id2idx, dict) else None
+                self.edge_features[etype] = Feature(
+                    feat, split_ratio, device_group_list, device, with_gpu,
+                    dtype=dtype, id2index=i2i)
+        elif edge_feature_data is not None:
+            self.edge_features = Feature(
+                edge_feature_data, split_ratio, device_group_list, device,
+                with_gpu, dtype=dtype, id2index=id2idx)
+        return self
+
+    def init_node_labels(self, node_label_data=None):
+        if node_label_data is not None:
+            self.node_labels = node_label_data
+        return self
+
+    def random_node_split(self, num_val: Union[int, float],
+                          num_test: Union[int, float],
+                          ntype: Optional[NodeType] = None):
+        n = self.num_nodes(ntype)
+        perm = torch.randperm(n)
+        nv = int(n * num_val) if isinstance(num_val, float) else num_val
+        nt = int(n * num_test) if isinstance(num_test, float) else num_test
+        self.val_idx = perm[:nv]
+        self.test_idx = perm[nv:nv + nt]
+        self.train_idx = perm[nv + nt:]
+        return self
+
+    # -- accessors ----------------------------------------------------------
+    def num_nodes(self, ntype: Optional[NodeType] = None) -> int:
+        if isinstance(self.graph, dict):
+            best = 0
+            for (src, _, dst), g in self.graph.items():
+                if ntype is None or dst == ntype or src == ntype:
+                    best = max(best, g.num_nodes)
+            return best
+        return self.graph.num_nodes if self.graph is not None else 0
+
+    def get_graph(self, etype: Optional[EdgeType] = None):
+        if isinstance(self.graph, dict):
+            return self.graph.get(etype) if etype is not None else self.graph
+        return self.graph
+
+    def get_node_feature(self, ntype: Optional[NodeType] = None):
+        if isinstance(self.node_features, dict):
+            return self.node_features.get(ntype)
+        return self.node_features
+
+    def get_edge_feature(self, etype: Optional[EdgeType] = None):
+        if isinstance(self.edge_features, dict):
+            return self.edge_features.get(etype)
+        return self.edge_features
+
+    def get_node_label(self, ntype: Optional[NodeType] = None):
+        if isinstance(self.node_labels, dict):
+            return self.node_labels.get(ntype)
+        return self.node_labels
+
+    # -- process sharing ----------------------------------------------------
+    def share_ipc(self):
+        for t in (self.node_labels, self.train_idx, self.val_idx,
+                  self.test_idx):
+            if torch.is_tensor(t) and not t.is_cuda:
+                t.share_memory_()
+        if isinstance(self.node_labels, dict):
+            for t in self.node_labels.values():
+                t.share_memory_()
+        return self
+
+    def __repr__(self):
+        return (f"Dataset(graph={self.graph}, "
+                f"node_features={self.node_features is not None})")

@@ -1,0 +1,165 @@
+"""Feature store: hot rows in HBM3E (optionally sharded across the 8 xGMI
+peers of a DeviceGroup), cold rows in pinned host memory served zero-copy.
+
+Design parity: reference python/data/feature.py (DeviceGroup :32-45, split
+logic :102-270) + csrc/cuda/unified_tensor.cu.  MI355X-native differences:
+ - the split point defaults from free HBM (288 GB/GPU usually fits the whole
+   feature matrix: split_ratio=1.0 means fully device-resident);
+ - the CPU part is a device-mapped view of pinned host memory, fed to the
+   same UnifiedFeatureStore gather kernel as the device segments;
+ - a DeviceGroup is the whole xGMI node (all-to-all links), not an NVLink
+   island.
+"""
+from typing import List, Optional, Union
+
+import torch
+
+
+class DeviceGroup:
+    """A set of GPUs whose HBM acts as one feature cache (xGMI all-to-all)."""
+
+    def __init__(self, group_id: int, device_list: List[int]):
+        self.group_id = group_id
+        self.device_list = list(device_list)
+
+    @property
+    def size(self):
+        return len(self.device_list)
+
+    def __repr__(self):
+        return f"DeviceGroup({self.group_id}, {self.device_list})"
+
+
+class Feature:
+    """2-D feature matrix with tiered storage.
+
+    Args:
+      feature_tensor: [N, F] host tensor (any dtype).
+      split_ratio: fraction of rows resident in GPU HBM (0..1). The hot rows
+        are rows [0, split) — reorder with `sort_by_in_degree` so hot ids
+        land first and pass `id2index`.
+      device_group_list: GPUs to shard the hot rows across (defaults to the
+        current device only).
+      device: device used for gathers.
+      with_gpu: force CPU-only when False.
+      dtype: optional cast.
+      id2index: optional [max_id+1] map from global ids to row positions
+        (set when features were reordered or cached).
+    """
+
+    def __init__(self, feature_tensor: torch.Tensor, split_ratio: float = 1.0,
+                 device_group_list: Optional[List[DeviceGroup]] = None,
+                 device: Optional[int] = None, with_gpu: bool = True,
+                 dtype: Optional[torch.dtype] = None,
+                 id2index: Optional[torch.Tensor] = None):
+        assert feature_tensor.dim() == 2
+        if dtype is not None and feature_tensor.dtype != dtype:
+            feature_tensor = feature_tensor.to(dtype)
+        self.cpu_tensor = feature_tensor.contiguous()
+        self.split_ratio = float(split_ratio)
+        self.device_group_list = device_group_list
+        self.device = device
+        self.with_gpu = with_gpu and torch.cuda.is_available()
+        self.id2index = id2index
+        self._store = None
+        self._lazy_done = False
+        self._device_rows = 0
+
+    # -- lazy device init ---------------------------------------------------
+    def lazy_init(self):
+        if self._lazy_done:
+            return
+        self._lazy_done = True
+        if not self.with_gpu:
+            return
+        from .. import _C
+
+        dev = self.device if self.device is not None else \
+            torch.cuda.current_device()
+        n = self.cpu_tensor.size(0)
+        hot = int(n * min(max(self.split_ratio, 0.0), 1.0))
+        store = _C.UnifiedFeatureStore(dev)
+        groups = self.device_group_list
+        if hot > 0:
+            if groups:
+                # Shard hot rows across the group's devices (xGMI peers).
+                devs = groups[0].device_list
+                per = (hot + len(devs) - 1) // len(devs)
+                start = 0
+                for d in devs:
+                    end = min(start + per, hot)
+                    if end <= start:
+                        break
+                    seg = self.cpu_tensor[start:end].to(
+                        torch.device("cuda", d), non_blocking=False)
+                    if d != dev:
+                        _C.enable_peer_access(dev, d)
+                    store.append(seg)
+                    self._keepalive = getattr(self, "_keepalive", [])
+                    self._keepalive.append(seg)
+                    start = end
+            else:
+                seg = self.cpu_tensor[:hot].to(torch.device("cuda", dev))
+                store.append(seg)
+                self._keepalive = [seg]
+        if hot < n:
+            cold = self.cpu_tensor[hot:]
+            if not cold.is_contiguous():
+                cold = cold.contiguous()
+            mapped = _C.host_mapped_view(cold, dev)
+            store.append(mapped)
+            self._cold_keepalive = (cold, mapped)
+        self._store = store
+        self._device_rows = hot
+
+    # -- lookups ------------------------------------------------------------
+    def __getitem__(self, ids: torch.Tensor) -> torch.Tensor:
+        rows = ids.long()
+        if self.id2index is not None:
+            rows = self.id2index.to(rows.device)[rows]
+        if self.with_gpu:
+            self.lazy_init()
+            dev = self.device if self.device is not None else \
+                torch.cuda.current_device()
+            rows = rows.to(torch.device("cuda", dev), non_blocking=True)
+            return self._store.gather(rows)
+        return self.cpu_tensor[rows.cpu()]
+
+    def cpu_get(self, ids: torch.Tensor) -> torch.Tensor:
+        """Host-side lookup (serves remote RPC feature requests)."""
+        rows = ids.long().cpu()
+        if self.id2index is not None:
+            rows = self.id2index.cpu()[rows]
+        return self.cpu_tensor[rows]
+
+    @property
+    def shape(self):
+        return self.cpu_tensor.shape
+
+    def size(self, dim: int):
+        return self.cpu_tensor.size(dim)
+
+    @property
+    def dtype(self):
+        return self.cpu_tensor.dtype
+
+    @property
+    def device_rows(self):
+        return self._device_rows
+
+    def share_ipc(self):
+        self.cpu_tensor.share_memory_()
+        if self.id2index is not None:
+            self.id2index.share_memory_()
+        return (self.cpu_tensor, self.split_ratio, self.device_group_list,
+                self.device, self.with_gpu, self.id2index)
+
+    @classmethod
+    def from_ipc(cls, handle):
+        cpu_tensor, split_ratio, groups, device, with_gpu, id2index = handle
+        f = cls(cpu_tensor, split_ratio, groups, device, with_gpu,
+                id2index=id2index)
+        return f
+
+    def __reduce__(self):
+        return (Feature.from_ipc, (self.share_ipc(),))

@@ -1,0 +1,208 @@
+"""Topology + Graph store.
+
+Design parity: reference python/data/graph.py (Topology :28-181, Graph
+:219-306) — but the device story is MI355X-native: there is no C++ Graph
+object carrying raw pointers.  A Graph materializes its CSR tensors either
+on the GPU ('CUDA' mode, HBM3E-resident), as pinned-host *device-mapped
+views* ('ZERO_COPY' mode — hipHostRegister + hipHostGetDevicePointer through
+glt_amd._C.host_mapped_view, reads ride UVA), or stays on CPU.  Every
+sampler kernel just sees device tensors.
+"""
+from typing import Optional, Tuple, Union
+
+import torch
+
+from ..utils.topo import coo_to_csc, coo_to_csr
+
+
+class Topology:
+    """Layout-normalizing CSR/CSC/COO container.
+
+    Args:
+      edge_index: [2, E] COO tensor (row, col), or a (indptr, indices) tuple
+        when input_layout is 'CSR'/'CSC'.
+      edge_ids: optional [E] global edge ids (default: COO position).
+      edge_weights: optional [E] float weights.
+      input_layout: 'COO' | 'CSR' | 'CSC'.
+      layout: target layout, 'CSR' (edge_dir='out') or 'CSC' (edge_dir='in').
+    """
+
+    def __init__(self, edge_index, edge_ids: Optional[torch.Tensor] = None,
+                 edge_weights: Optional[torch.Tensor] = None,
+                 input_layout: str = "COO", layout: str = "CSR",
+                 num_nodes: Optional[int] = None):
+        input_layout = input_layout.upper()
+        layout = layout.upper()
+        assert layout in ("CSR", "CSC")
+        self.layout = layout
+
+        if input_layout == "COO":
+            if torch.is_tensor(edge_index):
+                row, col = edge_index[0], edge_index[1]
+            else:
+                row, col = edge_index
+            n = num_nodes
+            if n is None and row.numel() > 0:
+                n = int(max(int(row.max()), int(col.max()))) + 1
+            if layout == "CSR":
+                indptr, indices, eids, ew = coo_to_csr(
+                    row, col, edge_ids, edge_weights, num_rows=n)
+            else:
+                indptr, indices, eids, ew = coo_to_csc(
+                    row, col, edge_ids, edge_weights, num_cols=n)
+        elif input_layout in ("CSR", "CSC"):
+            indptr, indices = edge_index
+            eids = edge_ids
+            if eids is None:
+                eids = torch.arange(indices.numel(), dtype=torch.long)
+            ew = edge_weights
+            if input_layout != layout:
+                # convert via COO round trip
+                num_rows = indptr.numel() - 1
+                rows = torch.repeat_interleave(
+                    torch.arange(num_rows), indptr[1:] - indptr[:-1])
+                if layout == "CSC":
+                    indptr, indices, eids, ew = coo_to_csc(
+                        rows, indices, eids, ew, num_cols=num_nodes or num_rows)
+                else:
+                    indptr, indices, eids, ew = coo_to_csr(
+                        indices, rows, eids, ew, num_rows=num_nodes or num_rows)
+        else:
+            raise ValueError(f"unknown input_layout {input_layout}")
+
+        self.indptr = indptr.long().contiguous()
+        self.indices = indices.long().contiguous()
+        self.edge_ids = eids.long().contiguous() if eids is not None else None
+        self.edge_weights = (ew.float().contiguous()
+                             if ew is not None else None)
+
+    @property
+    def num_nodes(self) -> int:
+        return self.indptr.numel() - 1
+
+    @property
+    def num_edges(self) -> int:
+        return self.indices.numel()
+
+    @property
+    def degrees(self) -> torch.Tensor:
+        return self.indptr[1:] - self.indptr[:-1]
+
+    def to_coo(self) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+        rows = torch.repeat_interleave(
+            torch.arange(self.num_nodes, device=self.indptr.device),
+            self.degrees)
+        return rows, self.indices, self.edge_ids
+
+    def share_memory_(self):
+        for t in (self.indptr, self.indices, self.edge_ids,
+                  self.edge_weights):
+            if t is not None and not t.is_cuda:
+                t.share_memory_()
+        return self
+
+
+class Graph:
+    """Device-resident graph for sampling.
+
+    mode:
+      'CPU'       - host tensors, CPU sampler.
+      'ZERO_COPY' - pinned-host tensors mapped into the GPU address space
+                    (UVA); GPU kernels read over PCIe. Best for graphs larger
+                    than spare HBM.
+      'CUDA'      - CSR copied into HBM3E (288 GB/GPU fits ogbn-scale graphs
+                    outright; the default for performance).
+    """
+
+    def __init__(self, topo: Topology, mode: str = "ZERO_COPY",
+                 device: Optional[int] = None):
+        self.topo = topo
+        self.mode = mode.upper()
+        if self.mode == "DMA":  # reference naming
+            self.mode = "CUDA"
+        self.device = device
+        self._indptr = None
+        self._indices = None
+        self._edge_ids = None
+        self._edge_weights = None
+        self._lazy_done = False
+
+    # -- device materialization -------------------------------------------
+    def lazy_init(self):
+        if self._lazy_done:
+            return
+        topo = self.topo
+        if self.mode == "CPU" or not torch.cuda.is_available():
+            self._indptr = topo.indptr
+            self._indices = topo.indices
+            self._edge_ids = topo.edge_ids
+            self._edge_weights = topo.edge_weights
+        elif self.mode == "CUDA":
+            dev = torch.device("cuda",
+                               self.device if self.device is not None else
+                               torch.cuda.current_device())
+            self._indptr = topo.indptr.to(dev)
+            self._indices = topo.indices.to(dev)
+            self._edge_ids = (topo.edge_ids.to(dev)
+                              if topo.edge_ids is not None else None)
+            self._edge_weights = (topo.edge_weights.to(dev)
+                                  if topo.edge_weights is not None else None)
+        elif self.mode == "ZERO_COPY":
+            from .. import _C
+
+            dev = self.device if self.device is not None else \
+                torch.cuda.current_device()
+
+            def mapped(t):
+                return None if t is None else _C.host_mapped_view(t, dev)
+
+            self._indptr = mapped(topo.indptr)
+            self._indices = mapped(topo.indices)
+            self._edge_ids = mapped(topo.edge_ids)
+            self._edge_weights = mapped(topo.edge_weights)
+        else:
+            raise ValueError(f"unknown graph mode {self.mode}")
+        self._lazy_done = True
+
+    @property
+    def indptr(self):
+        self.lazy_init()
+        return self._indptr
+
+    @property
+    def indices(self):
+        self.lazy_init()
+        return self._indices
+
+    @property
+    def edge_ids(self):
+        self.lazy_init()
+        return self._edge_ids
+
+    @property
+    def edge_weights(self):
+        self.lazy_init()
+        return self._edge_weights
+
+    @property
+    def num_nodes(self):
+        return self.topo.num_nodes
+
+    @property
+    def num_edges(self):
+        return self.topo.num_edges
+
+    def share_ipc(self):
+        """Make the host topology shareable across processes (fork/spawn via
+        torch.multiprocessing shared memory); device modes re-materialize
+        lazily in the child."""
+        self.topo.share_memory_()
+        return (self.topo, self.mode, self.device)
+
+    @classmethod
+    def from_ipc(cls, handle):
+        topo, mode, device = handle
+        return cls(topo, mode, device)
+
+    def __reduce__(self):
+        return (Graph.from_ipc, (self.share_ipc(),))

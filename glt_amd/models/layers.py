@@ -1,0 +1,116 @@
+"""GNN conv layers over glt_amd sampled batches.
+
+The reference delegates modeling to PyG (reference README.md:279-303);
+this image has no PyG, so glt_amd provides the conv layers its examples
+and benchmarks need, written against the glt_amd batch convention:
+``edge_index[0]`` = target-side (seed) local index, ``edge_index[1]`` =
+source/neighbor local index; aggregation flows 1 -> 0.
+
+All dense math lands on rocBLAS/hipBLASLt through torch.nn.Linear; the
+sparse aggregation uses index_add_ (rocPRIM-backed scatter) — a fused
+CSR-segment kernel is a planned optimization.
+"""
+import math
+from typing import Optional, Tuple
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+def _degree(target: torch.Tensor, num_nodes: int) -> torch.Tensor:
+    return torch.bincount(target, minlength=num_nodes).clamp_(min=1)
+
+
+class SAGEConv(nn.Module):
+    """GraphSAGE mean aggregator."""
+
+    def __init__(self, in_channels: int, out_channels: int,
+                 root_weight: bool = True, bias: bool = True):
+        super().__init__()
+        self.lin_l = nn.Linear(in_channels, out_channels, bias=bias)
+        self.lin_r = (nn.Linear(in_channels, out_channels, bias=False)
+                      if root_weight else None)
+
+    def forward(self, x: torch.Tensor,
+                edge_index: torch.Tensor) -> torch.Tensor:
+        n = x.size(0)
+        tgt, src = edge_index[0], edge_index[1]
+        agg = torch.zeros_like(x)
+        agg.index_add_(0, tgt, x.index_select(0, src))
+        agg = agg / _degree(tgt, n).unsqueeze(1).to(x.dtype)
+        out = self.lin_l(agg)
+        if self.lin_r is not None:
+            out = out + self.lin_r(x)
+        return out
+
+
+class GCNConv(nn.Module):
+    """GCN with symmetric degree normalization (computed on the sampled
+    subgraph)."""
+
+    def __init__(self, in_channels: int, out_channels: int,
+                 bias: bool = True):
+        super().__init__()
+        self.lin = nn.Linear(in_channels, out_channels, bias=bias)
+
+    def forward(self, x, edge_index):
+        n = x.size(0)
+        tgt, src = edge_index[0], edge_index[1]
+        deg = _degree(torch.cat([tgt, src]), n).to(x.dtype)
+        norm = deg.rsqrt()
+        h = self.lin(x)
+        msg = h.index_select(0, src) * norm[src].unsqueeze(1)
+        out = torch.zeros_like(h)
+        out.index_add_(0, tgt, msg)
+        return out * norm.unsqueeze(1)
+
+
+class GATConv(nn.Module):
+    """Multi-head graph attention (GATv1-style, scatter softmax)."""
+
+    def __init__(self, in_channels: int, out_channels: int, heads: int = 1,
+                 concat: bool = True, negative_slope: float = 0.2,
+                 dropout: float = 0.0, bias: bool = True):
+        super().__init__()
+        self.heads = heads
+        self.out_channels = out_channels
+        self.concat = concat
+        self.negative_slope = negative_slope
+        self.dropout = dropout
+        self.lin = nn.Linear(in_channels, heads * out_channels, bias=False)
+        self.att_src = nn.Parameter(torch.empty(1, heads, out_channels))
+        self.att_dst = nn.Parameter(torch.empty(1, heads, out_channels))
+        self.bias = nn.Parameter(torch.zeros(
+            heads * out_channels if concat else out_channels)) if bias \
+            else None
+        nn.init.xavier_uniform_(self.att_src)
+        nn.init.xavier_uniform_(self.att_dst)
+
+    def forward(self, x, edge_index):
+        n = x.size(0)
+        h = self.lin(x).view(n, self.heads, self.out_channels)
+        tgt, src = edge_index[0], edge_index[1]
+        alpha_src = (h * self.att_src).sum(-1)  # [n, H]
+        alpha_dst = (h * self.att_dst).sum(-1)
+        e = alpha_dst[tgt] + alpha_src[src]  # [E, H]
+        e = F.leaky_relu(e, self.negative_slope)
+        # scatter softmax over tgt
+        e_max = torch.full((n, self.heads), float("-inf"),
+                           device=e.device, dtype=e.dtype)
+        e_max.scatter_reduce_(0, tgt.unsqueeze(1).expand_as(e), e,
+                              reduce="amax", include_self=True)
+        e = (e - e_max[tgt]).exp()
+        denom = torch.zeros(n, self.heads, device=e.device, dtype=e.dtype)
+        denom.index_add_(0, tgt, e)
+        alpha = e / denom.clamp(min=1e-16)[tgt]
+        if self.training and self.dropout > 0:
+            alpha = F.dropout(alpha, p=self.dropout)
+        msg = h.index_select(0, src) * alpha.unsqueeze(-1)
+        out = torch.zeros_like(h)
+        out.index_add_(0, tgt, msg)
+        out = out.reshape(n, self.heads * self.out_channels) if self.concat \
+            else out.mean(dim=1)
+        if self.bias is not None:
+            out = out + self.bias
+        return out

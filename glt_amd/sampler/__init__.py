@@ -1,0 +1,13 @@
+from .base import (BaseSampler, EdgeIndex, EdgeSamplerInput,
+                   HeteroSamplerOutput, NegativeSampling,
+                   NegativeSamplingMode, NeighborOutput, NodeSamplerInput,
+                   SamplerOutput, SamplingConfig, SamplingType)
+from .neighbor_sampler import NeighborSampler
+from .negative_sampler import RandomNegativeSampler
+
+__all__ = [
+    "BaseSampler", "EdgeIndex", "EdgeSamplerInput", "HeteroSamplerOutput",
+    "NegativeSampling", "NegativeSamplingMode", "NeighborOutput",
+    "NodeSamplerInput", "SamplerOutput", "SamplingConfig", "SamplingType",
+    "NeighborSampler", "RandomNegativeSampler",
+]

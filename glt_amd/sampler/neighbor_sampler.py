@@ -1,0 +1,434 @@
+"""NeighborSampler — the local multi-hop sampling engine.
+
+Capability parity: reference python/sampler/neighbor_sampler.py (homo &
+hetero multi-hop :169-317, sample_from_edges with binary/triplet negatives
+:319-446, subgraph :474-498, sample_prob :500-627).  Fresh implementation:
+one code path for CPU and GPU (native ops auto-dispatch on tensor device),
+hetero induction orchestrated in python over per-type native inducers,
+row expansion via torch.repeat_interleave instead of bespoke kernels.
+"""
+import threading
+from typing import Dict, List, Optional, Union
+
+import torch
+
+from ..data import Graph
+from ..typing import EdgeType, NodeType, reverse_edge_type
+from ..utils.tracing import trace_region
+from .base import (BaseSampler, EdgeSamplerInput, HeteroSamplerOutput,
+                   NegativeSampling, NeighborOutput, NodeSamplerInput,
+                   SamplerOutput)
+
+
+def _relabel(uniq: torch.Tensor, ids: torch.Tensor) -> torch.Tensor:
+    """Positions of `ids` within insertion-ordered unique list `uniq`."""
+    sorted_uniq, perm = torch.sort(uniq)
+    pos = torch.searchsorted(sorted_uniq, ids)
+    return perm[pos]
+
+
+class _PyHeteroInducer:
+    """Device-agnostic hetero inducer over per-type native inducers."""
+
+    def __init__(self, make_inducer):
+        self._make = make_inducer
+        self._inducers: Dict[NodeType, object] = {}
+        self._uniq: Dict[NodeType, torch.Tensor] = {}
+
+    def _get(self, ntype):
+        if ntype not in self._inducers:
+            self._inducers[ntype] = self._make()
+        return self._inducers[ntype]
+
+    def init_node(self, seed_dict: Dict[NodeType, torch.Tensor]):
+        self._inducers = {}
+        self._uniq = {}
+        out = {}
+        for ntype, seeds in seed_dict.items():
+            ind = self._get(ntype)
+            uniq = ind.init_node(seeds)
+            self._uniq[ntype] = uniq
+            out[ntype] = uniq
+        return out
+
+    def insert(self, ntype, ids: torch.Tensor) -> torch.Tensor:
+        ind = self._get(ntype)
+        if ntype not in self._uniq:
+            fresh = ind.init_node(ids)
+        elif hasattr(ind, "insert"):
+            fresh = ind.insert(ids)
+        else:  # CPUInducer path: emulate insert via induce_next with dummy src
+            fresh, _, _ = ind.induce_next(
+                torch.empty(0, dtype=torch.long), ids,
+                torch.empty(0, dtype=torch.long))
+        self._uniq[ntype] = (torch.cat([self._uniq[ntype], fresh])
+                             if ntype in self._uniq else fresh)
+        return fresh
+
+    def lookup(self, ntype, ids: torch.Tensor) -> torch.Tensor:
+        ind = self._inducers[ntype]
+        if hasattr(ind, "lookup"):
+            return ind.lookup(ids)
+        return _relabel(self._uniq[ntype], ids)
+
+    def nodes(self, ntype):
+        return self._uniq.get(ntype)
+
+
+class NeighborSampler(BaseSampler):
+    """Multi-hop neighbor sampler over a Graph (homo) or Dict[EdgeType,
+    Graph] (hetero)."""
+
+    def __init__(self, graph: Union[Graph, Dict[EdgeType, Graph]],
+                 num_neighbors: Optional[List[int]] = None,
+                 device: Optional[torch.device] = None,
+                 with_edge: bool = False, with_neg: bool = False,
+                 with_weight: bool = False, edge_dir: str = "out",
+                 seed: Optional[int] = None):
+        from .. import _C
+
+        self._C = _C
+        self.graph = graph
+        self.num_neighbors = num_neighbors
+        self.with_edge = with_edge
+        self.with_neg = with_neg
+        self.with_weight = with_weight
+        self.edge_dir = edge_dir
+        self.is_hetero = isinstance(graph, dict)
+        self._lock = threading.Lock()
+        if seed is not None:
+            _C.manual_seed(seed)
+        g0 = next(iter(graph.values())) if self.is_hetero else graph
+        if device is not None:
+            self.device = torch.device(device)
+        else:
+            # sample where the graph lives
+            self.device = g0.indptr.device
+        self._cpu_mode = self.device.type == "cpu"
+
+    # ------------------------------------------------------------------
+    def _make_inducer(self):
+        if self._cpu_mode:
+            return self._C.CPUInducer(1024)
+        return self._C.DeviceInducer(4096)
+
+    def _seeds_to_device(self, seeds: torch.Tensor) -> torch.Tensor:
+        return seeds.long().to(self.device, non_blocking=True)
+
+    def sample_one_hop(self, srcs: torch.Tensor, k: int,
+                       etype: Optional[EdgeType] = None) -> NeighborOutput:
+        """Uniform (or weighted) one-hop sample from srcs."""
+        g = self.graph[etype] if etype is not None else self.graph
+        weighted = self.with_weight and g.edge_weights is not None
+        nbrs, num, eids = self._C.sample_neighbors(
+            g.indptr, g.indices, srcs, k,
+            edge_ids=g.edge_ids if self.with_edge else None,
+            edge_weights=g.edge_weights if weighted else None,
+            with_edge=self.with_edge, weighted=weighted)
+        return NeighborOutput(nbrs, num, eids)
+
+    # -- homo ----------------------------------------------------------
+    def sample_from_nodes(self, inputs: NodeSamplerInput,
+                          **kwargs) -> Union[SamplerOutput,
+                                             HeteroSamplerOutput]:
+        if self.is_hetero:
+            return self._hetero_sample_from_nodes(inputs)
+        seeds = self._seeds_to_device(inputs.node)
+        with trace_region("sample_from_nodes"):
+            return self._sample_from_nodes(seeds,
+                                           metadata={"input_type": None})
+
+    def _sample_from_nodes(self, seeds: torch.Tensor,
+                           metadata=None) -> SamplerOutput:
+        inducer = self._make_inducer()
+        uniq_seeds = inducer.init_node(seeds)
+        out_nodes = [uniq_seeds]
+        num_nodes = [uniq_seeds.numel()]
+        num_edges = []
+        rows, cols, eids = [], [], []
+        srcs = uniq_seeds
+        for k in (self.num_neighbors or []):
+            out = self.sample_one_hop(srcs, k)
+            nodes, r, c = inducer.induce_next(srcs, out.nbr, out.nbr_num)
+            out_nodes.append(nodes)
+            num_nodes.append(nodes.numel())
+            num_edges.append(r.numel())
+            rows.append(r)
+            cols.append(c)
+            if out.edge is not None:
+                eids.append(out.edge)
+            srcs = nodes
+        node = torch.cat(out_nodes)
+        row = torch.cat(rows) if rows else torch.empty(
+            0, dtype=torch.long, device=self.device)
+        col = torch.cat(cols) if cols else torch.empty(
+            0, dtype=torch.long, device=self.device)
+        edge = torch.cat(eids) if eids else None
+        return SamplerOutput(
+            node=node, row=row, col=col, edge=edge, batch=uniq_seeds,
+            num_sampled_nodes=num_nodes, num_sampled_edges=num_edges,
+            device=self.device, metadata=metadata)
+
+    # -- hetero ---------------------------------------------------------
+    def _etype_fanout(self, etype: EdgeType, hop: int) -> int:
+        nn = self.num_neighbors
+        if isinstance(nn, dict):
+            fan = nn.get(etype)
+            if fan is None:
+                return 0
+            return fan[hop] if hop < len(fan) else 0
+        return nn[hop] if nn and hop < len(nn) else 0
+
+    def _num_hops(self) -> int:
+        nn = self.num_neighbors
+        if isinstance(nn, dict):
+            return max(len(v) for v in nn.values())
+        return len(nn or [])
+
+    def _hetero_sample_from_nodes(
+            self, inputs: NodeSamplerInput) -> HeteroSamplerOutput:
+        input_type = inputs.input_type
+        assert input_type is not None, "hetero sampling needs input_type"
+        seeds = self._seeds_to_device(inputs.node)
+        return self._hetero_multihop(
+            {input_type: seeds},
+            metadata={"input_type": input_type, "bs": seeds.numel()})
+
+    def _hetero_multihop(self, seed_dict: Dict[NodeType, torch.Tensor],
+                         metadata=None) -> HeteroSamplerOutput:
+        inducer = _PyHeteroInducer(self._make_inducer)
+        frontier = inducer.init_node(seed_dict)
+        out_nodes: Dict[NodeType, List[torch.Tensor]] = {
+            t: [v] for t, v in frontier.items()}
+        num_nodes = {t: [v.numel()] for t, v in frontier.items()}
+        rows: Dict[EdgeType, List[torch.Tensor]] = {}
+        cols: Dict[EdgeType, List[torch.Tensor]] = {}
+        eids: Dict[EdgeType, List[torch.Tensor]] = {}
+        num_edges: Dict[EdgeType, List[int]] = {}
+
+        for hop in range(self._num_hops()):
+            next_frontier: Dict[NodeType, List[torch.Tensor]] = {}
+            hop_results = []
+            for etype, g in self.graph.items():
+                # In 'out' mode an etype (src, rel, dst) is sampled from its
+                # src-type frontier; in 'in' mode graphs are CSC keyed the
+                # same but we walk from dst.
+                walk_from = etype[0] if self.edge_dir == "out" else etype[2]
+                srcs = frontier.get(walk_from)
+                if srcs is None or srcs.numel() == 0:
+                    continue
+                k = self._etype_fanout(etype, hop)
+                if k == 0:
+                    continue
+                out = self.sample_one_hop(srcs, k, etype=etype)
+                hop_results.append((etype, srcs, out))
+            # Phase 1: insert all new nodes (deterministic etype order).
+            for etype, srcs, out in hop_results:
+                into = etype[2] if self.edge_dir == "out" else etype[0]
+                fresh = inducer.insert(into, out.nbr)
+                out_nodes.setdefault(into, []).append(fresh)
+                num_nodes.setdefault(into, []).append(fresh.numel())
+                next_frontier.setdefault(into, []).append(fresh)
+            # Phase 2: relabel edges.  For edge_dir='in' the walk goes
+            # dst->src over CSC, so results are stored under the reversed
+            # edge type (reference neighbor_sampler.py:261-269).
+            for etype, srcs, out in hop_results:
+                src_t = etype[0] if self.edge_dir == "out" else etype[2]
+                dst_t = etype[2] if self.edge_dir == "out" else etype[0]
+                key = (etype if self.edge_dir == "out"
+                       else reverse_edge_type(etype))
+                src_local = inducer.lookup(src_t, srcs)
+                r = torch.repeat_interleave(src_local, out.nbr_num)
+                c = inducer.lookup(dst_t, out.nbr)
+                rows.setdefault(key, []).append(r)
+                cols.setdefault(key, []).append(c)
+                num_edges.setdefault(key, []).append(r.numel())
+                if out.edge is not None:
+                    eids.setdefault(key, []).append(out.edge)
+            frontier = {t: torch.cat(v) for t, v in next_frontier.items()
+                        if v}
+        node = {t: torch.cat(v) for t, v in out_nodes.items()}
+        row = {et: torch.cat(v) for et, v in rows.items()}
+        col = {et: torch.cat(v) for et, v in cols.items()}
+        edge = {et: torch.cat(v) for et, v in eids.items()} if eids else None
+        batch = {t: self._uniq_of(seed_dict, inducer, t)
+                 for t in seed_dict.keys()}
+        return HeteroSamplerOutput(
+            node=node, row=row, col=col, edge=edge, batch=batch,
+            num_sampled_nodes=num_nodes, num_sampled_edges=num_edges,
+            edge_types=list(self.graph.keys()),
+            input_type=(metadata or {}).get("input_type"),
+            device=self.device, metadata=metadata)
+
+    @staticmethod
+    def _uniq_of(seed_dict, inducer, t):
+        n = inducer.nodes(t)
+        # seeds of type t are the first len(unique seeds) entries
+        return n[: torch.unique(seed_dict[t]).numel()] if n is not None \
+            else seed_dict[t]
+
+    # -- link sampling ---------------------------------------------------
+    def sample_from_edges(self, inputs: EdgeSamplerInput,
+                          **kwargs) -> Union[SamplerOutput,
+                                             HeteroSamplerOutput]:
+        neg = inputs.neg_sampling
+        if self.is_hetero:
+            return self._hetero_sample_from_edges(inputs)
+        row = self._seeds_to_device(inputs.row)
+        col = self._seeds_to_device(inputs.col)
+        num_pos = row.numel()
+        g = self.graph
+        if neg is not None and neg.is_binary():
+            num_neg = int(num_pos * neg.amount)
+            neg_edges = self._C.sample_negative(
+                g.indptr, g.indices, g.num_nodes, num_neg, 5, True)
+            seeds = torch.cat([row, col, neg_edges[0], neg_edges[1]])
+            out = self._sample_from_nodes(seeds)
+            local = _relabel_through(self, out, seeds)
+            n_neg = neg_edges.size(1)
+            eli = torch.stack([
+                torch.cat([local[:num_pos], local[2 * num_pos:
+                                                  2 * num_pos + n_neg]]),
+                torch.cat([local[num_pos:2 * num_pos],
+                           local[2 * num_pos + n_neg:]]),
+            ])
+            label = torch.cat([
+                inputs.label.to(self.device) if inputs.label is not None
+                else torch.ones(num_pos, device=self.device),
+                torch.zeros(n_neg, device=self.device),
+            ])
+            out.metadata = {"edge_label_index": eli, "edge_label": label,
+                            "input_type": None}
+            return out
+        if neg is not None and neg.is_triplet():
+            amount = int(neg.amount)
+            neg_dst = self._C.sample_negative(
+                g.indptr, g.indices, g.num_nodes, num_pos * amount, 5,
+                True)[1]
+            seeds = torch.cat([row, col, neg_dst])
+            out = self._sample_from_nodes(seeds)
+            local = _relabel_through(self, out, seeds)
+            out.metadata = {
+                "src_index": local[:num_pos],
+                "dst_pos_index": local[num_pos:2 * num_pos],
+                "dst_neg_index":
+                    local[2 * num_pos:].view(num_pos, amount)
+                    if local.numel() >= 2 * num_pos else local[2 * num_pos:],
+                "input_type": None,
+            }
+            return out
+        seeds = torch.cat([row, col])
+        out = self._sample_from_nodes(seeds)
+        local = _relabel_through(self, out, seeds)
+        eli = torch.stack([local[:num_pos], local[num_pos:]])
+        label = (inputs.label.to(self.device)
+                 if inputs.label is not None else None)
+        out.metadata = {"edge_label_index": eli, "edge_label": label,
+                        "input_type": None}
+        return out
+
+    def _hetero_sample_from_edges(
+            self, inputs: EdgeSamplerInput) -> HeteroSamplerOutput:
+        etype = inputs.input_type
+        assert etype is not None
+        src_t, _, dst_t = etype
+        row = self._seeds_to_device(inputs.row)
+        col = self._seeds_to_device(inputs.col)
+        num_pos = row.numel()
+        neg = inputs.neg_sampling
+        neg_dst = None
+        if neg is not None:
+            g = self.graph[etype]
+            amount = (int(num_pos * neg.amount) if neg.is_binary()
+                      else num_pos * int(neg.amount))
+            neg_pair = self._C.sample_negative(
+                g.indptr, g.indices, g.num_nodes, amount, 5, True)
+            neg_dst = neg_pair[1]
+        if src_t == dst_t:
+            seeds = {src_t: torch.cat([row, col] +
+                                      ([neg_dst] if neg_dst is not None
+                                       else []))}
+        else:
+            dst_seeds = torch.cat([col] + ([neg_dst]
+                                           if neg_dst is not None else []))
+            seeds = {src_t: row, dst_t: dst_seeds}
+        out = self._hetero_multihop(seeds, metadata={"input_type": etype})
+        src_local = _relabel(out.node[src_t], row)
+        dst_local = _relabel(out.node[dst_t], col)
+        if neg is not None and neg.is_triplet():
+            out.metadata.update({
+                "src_index": src_local,
+                "dst_pos_index": dst_local,
+                "dst_neg_index": _relabel(out.node[dst_t],
+                                          neg_dst).view(num_pos, -1),
+            })
+        else:
+            if neg_dst is not None:
+                neg_src = row.repeat(
+                    (neg_dst.numel() + num_pos - 1) // num_pos)[
+                        : neg_dst.numel()]
+                eli = torch.stack([
+                    torch.cat([src_local,
+                               _relabel(out.node[src_t], neg_src)]),
+                    torch.cat([dst_local,
+                               _relabel(out.node[dst_t], neg_dst)]),
+                ])
+                label = torch.cat([
+                    torch.ones(num_pos, device=self.device),
+                    torch.zeros(neg_dst.numel(), device=self.device)])
+            else:
+                eli = torch.stack([src_local, dst_local])
+                label = (inputs.label.to(self.device)
+                         if inputs.label is not None else None)
+            out.metadata.update({"edge_label_index": eli,
+                                 "edge_label": label})
+        return out
+
+    # -- subgraph ---------------------------------------------------------
+    def subgraph(self, inputs: NodeSamplerInput) -> SamplerOutput:
+        """Induce the full edge set among the (optionally multi-hop
+        expanded) seed set (SamplingType.SUBGRAPH; e.g. SEAL)."""
+        seeds = self._seeds_to_device(inputs.node)
+        g = self.graph
+        nodes = seeds
+        for k in (self.num_neighbors or []):
+            out = self.sample_one_hop(nodes, k)
+            nodes = torch.cat([nodes, out.nbr])
+        uniq, rows, cols, eids = self._C.node_subgraph(
+            g.indptr, g.indices, nodes, edge_ids=g.edge_ids,
+            with_edge=self.with_edge)
+        return SamplerOutput(
+            node=uniq, row=rows, col=cols, edge=eids, device=self.device,
+            metadata={"batch_size": seeds.numel(), "input_type": None})
+
+    # -- random walk -------------------------------------------------------
+    def random_walk(self, seeds: torch.Tensor, walk_len: int) -> torch.Tensor:
+        g = self.graph
+        return self._C.random_walk(g.indptr, g.indices,
+                                   self._seeds_to_device(seeds), walk_len)
+
+    # -- importance probability (trim) --------------------------------------
+    def sample_prob(self, seeds: torch.Tensor,
+                    num_nodes: Optional[int] = None) -> torch.Tensor:
+        """Per-node inclusion probability after the configured multi-hop
+        sampling from `seeds` (reference sample_prob,
+        neighbor_sampler.py:500-627)."""
+        g = self.graph
+        n = num_nodes or g.num_nodes
+        seeds = self._seeds_to_device(seeds)
+        prob = torch.zeros(n, dtype=torch.float32, device=self.device)
+        prob[seeds] = 1.0
+        frontier = seeds
+        for k in (self.num_neighbors or []):
+            prob = self._C.cal_nbr_prob(g.indptr, g.indices, prob, frontier,
+                                        k)
+            frontier = torch.nonzero(prob > 0).flatten()
+        return prob
+
+
+def _relabel_through(sampler: NeighborSampler, out: SamplerOutput,
+                     seeds: torch.Tensor) -> torch.Tensor:
+    """Local positions of `seeds` in out.node (insertion-ordered)."""
+    n_seed_nodes = out.num_sampled_nodes[0] if out.num_sampled_nodes else \
+        out.node.numel()
+    return _relabel(out.node[:n_seed_nodes], seeds)

@@ -1,0 +1,183 @@
+"""CPU sampler / inducer / subgraph / stitch unit tests on closed-form
+fixtures (reference test strategy: SURVEY.md §4)."""
+import collections
+
+import pytest
+import torch
+
+import glt_amd
+from glt_amd import _C
+from glt_amd.data import Graph, Topology
+
+
+def make_topo(ring_graph):
+    return Topology(ring_graph["edge_index"],
+                    num_nodes=ring_graph["num_nodes"])
+
+
+def test_topology_roundtrip(ring_graph):
+    topo = make_topo(ring_graph)
+    assert topo.num_nodes == 40
+    assert topo.num_edges == 80
+    assert (topo.degrees == 2).all()
+    rows, cols, eids = topo.to_coo()
+    # edges sorted per row; membership must match the input
+    orig = set(map(tuple, ring_graph["edge_index"].t().tolist()))
+    got = set(zip(rows.tolist(), cols.tolist()))
+    assert orig == got
+    # indices sorted within each row except wrap rows (39 -> 0,1 etc. are
+    # still ascending per row after sort)
+    for v in range(40):
+        s, e = topo.indptr[v].item(), topo.indptr[v + 1].item()
+        seg = topo.indices[s:e].tolist()
+        assert seg == sorted(seg)
+
+
+def test_sample_neighbors_full_and_partial(ring_graph):
+    topo = make_topo(ring_graph)
+    seeds = torch.tensor([0, 7, 39])
+    nbrs, num, eids = _C.sample_neighbors(topo.indptr, topo.indices, seeds,
+                                          -1, edge_ids=topo.edge_ids,
+                                          with_edge=True)
+    assert num.tolist() == [2, 2, 2]
+    got = set(zip([0, 0, 7, 7, 39, 39], nbrs.tolist()))
+    assert got == {(0, 1), (0, 2), (7, 8), (7, 9), (39, 0), (39, 1)}
+    # sampled edge ids must correspond: indices[pos] == nbr
+    flat = topo.indices[...]
+    for e, nb in zip(eids.tolist(), nbrs.tolist()):
+        # eids are the original COO positions; check col matches
+        assert ring_graph["edge_index"][1][e].item() == nb
+
+    # k=1: one distinct neighbor from each seed's true set
+    for trial in range(5):
+        nbrs1, num1, _ = _C.sample_neighbors(topo.indptr, topo.indices,
+                                             seeds, 1)
+        assert num1.tolist() == [1, 1, 1]
+        for s, nb in zip(seeds.tolist(), nbrs1.tolist()):
+            assert (nb - s) % 40 in (1, 2)
+
+
+def test_sample_without_replacement_distinct():
+    # star graph: node 0 -> 1..100
+    n = 101
+    rows = [0] * 100
+    cols = list(range(1, 101))
+    topo = Topology(torch.tensor([rows, cols]), num_nodes=n)
+    seeds = torch.tensor([0])
+    for k in (5, 50, 99):
+        nbrs, num, _ = _C.sample_neighbors(topo.indptr, topo.indices, seeds,
+                                           k)
+        assert num.item() == k
+        assert len(set(nbrs.tolist())) == k  # distinct == without replacement
+
+
+def test_sample_uniformity():
+    # node 0 with 20 neighbors, k=5 -> each neighbor ~25% per draw
+    glt_amd.seed_everything(7)
+    n = 21
+    topo = Topology(torch.tensor([[0] * 20, list(range(1, 21))]),
+                    num_nodes=n)
+    counts = collections.Counter()
+    trials = 3000
+    seeds = torch.zeros(trials, dtype=torch.long)  # 3000 copies of node 0
+    nbrs, num, _ = _C.sample_neighbors(topo.indptr, topo.indices, seeds, 5)
+    assert num.sum().item() == trials * 5
+    for nb in nbrs.tolist():
+        counts[nb] += 1
+    expected = trials * 5 / 20
+    for v in range(1, 21):
+        assert abs(counts[v] - expected) < expected * 0.25, counts
+
+
+def test_weighted_sampling_bias():
+    glt_amd.seed_everything(3)
+    # node 0 -> 1 (w=9), 2 (w=1): expect ~90/10 split
+    topo = Topology(torch.tensor([[0, 0], [1, 2]]),
+                    edge_weights=torch.tensor([9.0, 1.0]), num_nodes=3)
+    seeds = torch.zeros(2000, dtype=torch.long)
+    nbrs, num, _ = _C.sample_neighbors(topo.indptr, topo.indices, seeds, 1,
+                                       edge_weights=topo.edge_weights,
+                                       weighted=True)
+    frac = (nbrs == 1).float().mean().item()
+    assert 0.82 < frac < 0.97, frac
+
+
+def test_negative_sampler(ring_graph):
+    topo = make_topo(ring_graph)
+    neg = _C.sample_negative(topo.indptr, topo.indices, 40, 64, trials=10)
+    assert neg.size(0) == 2 and neg.size(1) > 0
+    for r, c in neg.t().tolist():
+        assert (c - r) % 40 not in (1, 2), (r, c)
+
+
+def test_random_walk(ring_graph):
+    topo = make_topo(ring_graph)
+    seeds = torch.arange(10)
+    walks = _C.random_walk(topo.indptr, topo.indices, seeds, 4)
+    assert walks.shape == (10, 5)
+    steps = (walks[:, 1:] - walks[:, :-1]) % 40
+    assert ((steps == 1) | (steps == 2)).all()
+
+
+def test_inducer_incremental():
+    ind = _C.CPUInducer(16)
+    uniq = ind.init_node(torch.tensor([3, 5, 3, 7]))
+    assert uniq.tolist() == [3, 5, 7]
+    nodes, rows, cols = ind.induce_next(
+        torch.tensor([3, 5]), torch.tensor([5, 9, 9, 11]),
+        torch.tensor([2, 2]))
+    assert nodes.tolist() == [9, 11]
+    assert rows.tolist() == [0, 0, 1, 1]
+    # cols relabeled: 5->1, 9->3, 11->4
+    assert cols.tolist() == [1, 3, 3, 4]
+
+
+def test_hetero_inducer():
+    ind = _C.CPUHeteroInducer(16)
+    uniq = ind.init_node({"user": torch.tensor([1, 2])})
+    assert uniq["user"].tolist() == [1, 2]
+    nodes, rows, cols = ind.induce_next(
+        ["user"], ["item"], [torch.tensor([1, 2])],
+        [torch.tensor([10, 11, 11, 12])], [torch.tensor([2, 2])])
+    assert nodes["item"].tolist() == [10, 11, 12]
+    assert rows[0].tolist() == [0, 0, 1, 1]
+    assert cols[0].tolist() == [0, 1, 1, 2]
+
+
+def test_node_subgraph(ring_graph):
+    topo = make_topo(ring_graph)
+    nodes = torch.tensor([0, 1, 2, 3])
+    uniq, rows, cols, eids = _C.node_subgraph(topo.indptr, topo.indices,
+                                              nodes,
+                                              edge_ids=topo.edge_ids,
+                                              with_edge=True)
+    assert uniq.tolist() == [0, 1, 2, 3]
+    got = set(zip(uniq[rows].tolist(), uniq[cols].tolist()))
+    assert got == {(0, 1), (0, 2), (1, 2), (1, 3), (2, 3)}
+
+
+def test_stitch(ring_graph):
+    # two partitions serve interleaved seed positions
+    idx0 = torch.tensor([0, 2])
+    idx1 = torch.tensor([1, 3])
+    nbrs0 = torch.tensor([10, 11, 20])
+    num0 = torch.tensor([2, 1])
+    nbrs1 = torch.tensor([30, 40, 41])
+    num1 = torch.tensor([1, 2])
+    nbrs, num, _ = _C.stitch_sample_results(4, [idx0, idx1], [nbrs0, nbrs1],
+                                            [num0, num1])
+    assert num.tolist() == [2, 1, 1, 2]
+    assert nbrs.tolist() == [10, 11, 30, 20, 40, 41]
+
+
+def test_cal_nbr_prob(ring_graph):
+    topo = make_topo(ring_graph)
+    prob = torch.zeros(40)
+    prob[0] = 1.0
+    out = _C.cal_nbr_prob(topo.indptr, topo.indices, prob,
+                          torch.tensor([0]), 1)
+    # node 0 has 2 nbrs, k=1 -> each kept with p=0.5
+    assert abs(out[1].item() - 0.5) < 1e-5
+    assert abs(out[2].item() - 0.5) < 1e-5
+    assert out[0].item() == 1.0  # seed stays
+    assert out[3].item() == 0.0
