@@ -14,6 +14,8 @@ import os
 # RCCL over xGMI: the host driver supports dmabuf IPC only.
 os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
 
+import torch  # noqa: F401  (loads libc10/libtorch before our extension)
+
 from . import _C  # native core (built in-tree; fail loudly if missing)
 from . import data, loader, models, partition, sampler, utils
 from .data import Dataset, DeviceGroup, Feature, Graph, Topology

@@ -174,6 +174,12 @@ class HIPInducer {
     return {nodes, rows, cols};
   }
 
+  // Insert ids (dedup against all previous hops); returns the new nodes.
+  torch::Tensor insert(const torch::Tensor& ids) {
+    ensure_capacity(count_ + ids.size(0), false);
+    return insert_and_assign(ids);
+  }
+
   // Relabel arbitrary global ids through the current table (-1 if absent).
   torch::Tensor lookup(const torch::Tensor& ids) {
     auto out = torch::empty_like(ids);
@@ -292,6 +298,9 @@ hip_inducer_induce_next(HIPInducer* ind, const torch::Tensor& srcs,
 }
 torch::Tensor hip_inducer_lookup(HIPInducer* ind, const torch::Tensor& ids) {
   return ind->lookup(ids);
+}
+torch::Tensor hip_inducer_insert(HIPInducer* ind, const torch::Tensor& ids) {
+  return ind->insert(ids);
 }
 int64_t hip_inducer_count(HIPInducer* ind) { return ind->count(); }
 

@@ -1,2 +1,14 @@
-"""Offline graph partitioning (filled in as the distributed runtime lands)."""
-__all__ = []
+from .base import (FeaturePartitionData, GraphPartitionData,
+                   PartitionerBase, cat_feature_cache, load_partition,
+                   save_meta)
+from .partition_book import (GLTPartitionBook, PartitionBook,
+                             RangePartitionBook)
+from .random_partitioner import RandomPartitioner
+from .frequency_partitioner import FrequencyPartitioner
+
+__all__ = [
+    "FeaturePartitionData", "GraphPartitionData", "PartitionerBase",
+    "cat_feature_cache", "load_partition", "save_meta", "GLTPartitionBook",
+    "PartitionBook", "RangePartitionBook", "RandomPartitioner",
+    "FrequencyPartitioner",
+]

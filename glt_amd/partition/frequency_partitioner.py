@@ -1,0 +1,81 @@
+"""Frequency (hotness-aware) partitioner (parity: reference
+python/partition/frequency_partitioner.py:103-205).
+
+Inputs per training partition: a node access-probability vector (e.g. from
+`NeighborSampler.sample_prob` over that partition's seeds).  Chunks of nodes
+are greedily assigned to the partition whose seeds access them most, and the
+hottest nodes of each partition are selected as its GPU feature cache.
+"""
+from typing import Dict, List, Optional, Union
+
+import torch
+
+from ..utils.common import parse_size
+from .base import PartitionerBase
+from .partition_book import GLTPartitionBook
+
+
+class FrequencyPartitioner(PartitionerBase):
+    def __init__(self, output_dir, num_parts, num_nodes, edge_index,
+                 probs: Union[List[torch.Tensor],
+                              Dict[str, List[torch.Tensor]]],
+                 node_feat=None, edge_feat=None, edge_weights=None,
+                 edge_assign_strategy: str = "by_src",
+                 chunk_size: int = 10_000_000,
+                 cache_memory_budget=None, cache_ratio: float = 0.0):
+        super().__init__(output_dir, num_parts, num_nodes, edge_index,
+                         node_feat, edge_feat, edge_weights,
+                         edge_assign_strategy, chunk_size)
+        self.probs = probs
+        self.cache_ratio = cache_ratio
+        self.cache_memory_budget = parse_size(cache_memory_budget)
+        self._cache_ids: Dict = {}
+
+    def _get_probs(self, ntype):
+        return self.probs[ntype] if isinstance(self.probs, dict) \
+            else self.probs
+
+    def _partition_node_ids(self, ntype=None):
+        n = (self.num_nodes[ntype] if isinstance(self.num_nodes, dict)
+             else self.num_nodes)
+        probs = self._get_probs(ntype)
+        assert len(probs) == self.num_parts
+        P = torch.stack([p[:n].float() for p in probs])  # [parts, n]
+        # Greedy chunk assignment balanced by current partition load.
+        chunk = 32
+        n_chunks = (n + chunk - 1) // chunk
+        node_pb = torch.zeros(n, dtype=torch.uint8)
+        loads = torch.zeros(self.num_parts)
+        # chunk score per partition = sum of probs in chunk
+        pad = n_chunks * chunk - n
+        Pp = torch.nn.functional.pad(P, (0, pad))
+        chunk_scores = Pp.view(self.num_parts, n_chunks, chunk).sum(-1)
+        order = torch.argsort(chunk_scores.max(0).values, descending=True)
+        cap = (n_chunks + self.num_parts - 1) // self.num_parts
+        counts = torch.zeros(self.num_parts, dtype=torch.long)
+        for c in order.tolist():
+            scores = chunk_scores[:, c].clone()
+            scores[counts >= cap] = float("-inf")
+            p = int(torch.argmax(scores))
+            s, e = c * chunk, min((c + 1) * chunk, n)
+            node_pb[s:e] = p
+            counts[p] += 1
+        ids_list = [torch.nonzero(node_pb == p).flatten()
+                    for p in range(self.num_parts)]
+        # hot-cache selection per partition
+        feat = self.node_feat if not isinstance(self.node_feat, dict) \
+            else (self.node_feat or {}).get(ntype)
+        for p in range(self.num_parts):
+            k = 0
+            if self.cache_memory_budget and feat is not None:
+                row_bytes = feat[0].numel() * feat.element_size()
+                k = int(self.cache_memory_budget // max(row_bytes, 1))
+            elif self.cache_ratio > 0:
+                k = int(n * self.cache_ratio)
+            if k > 0:
+                hot = torch.argsort(P[p], descending=True)[:k]
+                self._cache_ids[(ntype, p)] = hot
+        return ids_list, GLTPartitionBook(node_pb)
+
+    def _cache_node_ids(self, ntype, partition_idx):
+        return self._cache_ids.get((ntype, partition_idx))
