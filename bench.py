@@ -157,7 +157,8 @@ def main():
             it = iter(loader)
             data = next(it)
         opt.zero_grad(set_to_none=True)
-        out = model(data.x, data.edge_index)[:data.batch_size]
+        out = model(data.x, data.edge_index, data.num_sampled_nodes,
+                    data.num_sampled_edges)[:data.batch_size]
         loss = F.cross_entropy(out, data.y[:data.batch_size])
         loss.backward()
         opt.step()

@@ -5,6 +5,9 @@ import argparse
 import json
 import time
 
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import torch
 
 

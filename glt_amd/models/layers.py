@@ -32,16 +32,16 @@ class SAGEConv(nn.Module):
         self.lin_r = (nn.Linear(in_channels, out_channels, bias=False)
                       if root_weight else None)
 
-    def forward(self, x: torch.Tensor,
-                edge_index: torch.Tensor) -> torch.Tensor:
-        n = x.size(0)
+    def forward(self, x: torch.Tensor, edge_index: torch.Tensor,
+                num_target: int = None) -> torch.Tensor:
+        n = num_target if num_target is not None else x.size(0)
         tgt, src = edge_index[0], edge_index[1]
-        agg = torch.zeros_like(x)
+        agg = x.new_zeros(n, x.size(1))
         agg.index_add_(0, tgt, x.index_select(0, src))
         agg = agg / _degree(tgt, n).unsqueeze(1).to(x.dtype)
         out = self.lin_l(agg)
         if self.lin_r is not None:
-            out = out + self.lin_r(x)
+            out = out + self.lin_r(x[:n])
         return out
 
 
@@ -54,16 +54,17 @@ class GCNConv(nn.Module):
         super().__init__()
         self.lin = nn.Linear(in_channels, out_channels, bias=bias)
 
-    def forward(self, x, edge_index):
+    def forward(self, x, edge_index, num_target: int = None):
         n = x.size(0)
+        nt = num_target if num_target is not None else n
         tgt, src = edge_index[0], edge_index[1]
         deg = _degree(torch.cat([tgt, src]), n).to(x.dtype)
         norm = deg.rsqrt()
         h = self.lin(x)
         msg = h.index_select(0, src) * norm[src].unsqueeze(1)
-        out = torch.zeros_like(h)
+        out = h.new_zeros(nt, h.size(1))
         out.index_add_(0, tgt, msg)
-        return out * norm.unsqueeze(1)
+        return out * norm[:nt].unsqueeze(1)
 
 
 class GATConv(nn.Module):
@@ -87,8 +88,9 @@ class GATConv(nn.Module):
         nn.init.xavier_uniform_(self.att_src)
         nn.init.xavier_uniform_(self.att_dst)
 
-    def forward(self, x, edge_index):
+    def forward(self, x, edge_index, num_target: int = None):
         n = x.size(0)
+        nt = num_target if num_target is not None else n
         h = self.lin(x).view(n, self.heads, self.out_channels)
         tgt, src = edge_index[0], edge_index[1]
         alpha_src = (h * self.att_src).sum(-1)  # [n, H]
@@ -96,20 +98,20 @@ class GATConv(nn.Module):
         e = alpha_dst[tgt] + alpha_src[src]  # [E, H]
         e = F.leaky_relu(e, self.negative_slope)
         # scatter softmax over tgt
-        e_max = torch.full((n, self.heads), float("-inf"),
+        e_max = torch.full((nt, self.heads), float("-inf"),
                            device=e.device, dtype=e.dtype)
         e_max.scatter_reduce_(0, tgt.unsqueeze(1).expand_as(e), e,
                               reduce="amax", include_self=True)
         e = (e - e_max[tgt]).exp()
-        denom = torch.zeros(n, self.heads, device=e.device, dtype=e.dtype)
+        denom = torch.zeros(nt, self.heads, device=e.device, dtype=e.dtype)
         denom.index_add_(0, tgt, e)
         alpha = e / denom.clamp(min=1e-16)[tgt]
         if self.training and self.dropout > 0:
             alpha = F.dropout(alpha, p=self.dropout)
         msg = h.index_select(0, src) * alpha.unsqueeze(-1)
-        out = torch.zeros_like(h)
+        out = h.new_zeros(nt, self.heads, self.out_channels)
         out.index_add_(0, tgt, msg)
-        out = out.reshape(n, self.heads * self.out_channels) if self.concat \
+        out = out.reshape(nt, self.heads * self.out_channels) if self.concat \
             else out.mean(dim=1)
         if self.bias is not None:
             out = out + self.bias

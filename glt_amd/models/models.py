@@ -9,6 +9,26 @@ import torch.nn.functional as F
 from .layers import GATConv, GCNConv, SAGEConv
 
 
+def _layer_trim(num_sampled_nodes, num_sampled_edges, num_layers, layer):
+    """Rows/edges needed by layer `layer` (0-based) of an L-layer GNN over a
+    glt_amd multi-hop batch: layer l only needs nodes of hops
+    0..L-l and edges of hops 1..L-l (hop-ordered concatenation).
+    Returns (n_in_rows, n_edges, n_out_rows) or None when trim info is
+    unusable."""
+    if not num_sampled_nodes or not num_sampled_edges:
+        return None
+    nsn = [int(v) for v in num_sampled_nodes]
+    nse = [int(v) for v in num_sampled_edges]
+    L = num_layers
+    if len(nse) < L or len(nsn) < L + 1:
+        return None
+    keep_hops = L - layer          # edges of hops 1..keep_hops
+    n_edges = sum(nse[:keep_hops])
+    n_in = sum(nsn[:keep_hops + 1])
+    n_out = sum(nsn[:keep_hops])
+    return n_in, n_edges, n_out
+
+
 class GraphSAGE(nn.Module):
     def __init__(self, in_channels: int, hidden_channels: int,
                  num_layers: int, out_channels: Optional[int] = None,
@@ -22,9 +42,20 @@ class GraphSAGE(nn.Module):
             self.convs.append(SAGEConv(dims[i], dims[i + 1]))
         self.dropout = dropout
 
-    def forward(self, x, edge_index):
+    def forward(self, x, edge_index, num_sampled_nodes=None,
+                num_sampled_edges=None):
+        """Hop-wise trimmed forward: with the per-hop counts from a sampled
+        batch, layer l runs only over the rows/edges still reachable from
+        the seeds — ~(fan-out) x less compute on the deep layers."""
+        L = len(self.convs)
         for i, conv in enumerate(self.convs):
-            x = conv(x, edge_index)
+            trim = _layer_trim(num_sampled_nodes, num_sampled_edges, L, i)
+            if trim is not None:
+                n_in, n_edges, n_out = trim
+                x = conv(x[:n_in], edge_index[:, :n_edges],
+                         num_target=n_out)
+            else:
+                x = conv(x, edge_index)
             if i < len(self.convs) - 1:
                 x = F.relu(x)
                 x = F.dropout(x, p=self.dropout, training=self.training)

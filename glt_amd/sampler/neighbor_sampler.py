@@ -99,12 +99,13 @@ class NeighborSampler(BaseSampler):
         if seed is not None:
             _C.manual_seed(seed)
         g0 = next(iter(graph.values())) if self.is_hetero else graph
-        if device is not None:
-            self.device = torch.device(device)
-        else:
-            # sample where the graph lives
-            self.device = g0.indptr.device
-        self._cpu_mode = self.device.type == "cpu"
+        # Sampling always runs where the graph's CSR lives (device tensors
+        # for CUDA/ZERO_COPY modes, host for CPU mode); `device` only sets
+        # where outputs are reported.
+        self._sample_device = g0.indptr.device
+        self.device = torch.device(device) if device is not None \
+            else self._sample_device
+        self._cpu_mode = self._sample_device.type == "cpu"
 
     # ------------------------------------------------------------------
     def _make_inducer(self):
@@ -113,7 +114,7 @@ class NeighborSampler(BaseSampler):
         return self._C.DeviceInducer(4096)
 
     def _seeds_to_device(self, seeds: torch.Tensor) -> torch.Tensor:
-        return seeds.long().to(self.device, non_blocking=True)
+        return seeds.long().to(self._sample_device, non_blocking=True)
 
     def sample_one_hop(self, srcs: torch.Tensor, k: int,
                        etype: Optional[EdgeType] = None) -> NeighborOutput:

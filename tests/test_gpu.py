@@ -93,12 +93,14 @@ def test_device_inducer_growth():
     seeds = torch.arange(4).cuda()
     u = ind.init_node(seeds)
     assert u.cpu().tolist() == [0, 1, 2, 3]
+    seen = set(range(4))
     for hop in range(4):
         base = 4 * (hop + 1)
-        nbrs = torch.arange(base, base + 64).cuda()
-        fresh = ind.insert(nbrs)
-        assert fresh.cpu().tolist()[:4] == [base, base + 1, base + 2,
-                                            base + 3]
+        nbrs = list(range(base, base + 64))
+        fresh = ind.insert(torch.tensor(nbrs).cuda())
+        expected = [v for v in nbrs if v not in seen]
+        seen.update(nbrs)
+        assert fresh.cpu().tolist() == expected
     # lookup old ids still valid after growth
     lk = ind.lookup(torch.tensor([0, 1, 2, 3]).cuda())
     assert lk.cpu().tolist() == [0, 1, 2, 3]
