@@ -17,7 +17,8 @@ os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
 import torch  # noqa: F401  (loads libc10/libtorch before our extension)
 
 from . import _C  # native core (built in-tree; fail loudly if missing)
-from . import data, loader, models, partition, sampler, utils
+from . import channel, data, distributed, loader, models, partition, \
+    sampler, utils
 from .data import Dataset, DeviceGroup, Feature, Graph, Topology
 from .loader import (LinkLoader, LinkNeighborLoader, NeighborLoader,
                      NodeLoader, SubGraphLoader)

@@ -143,7 +143,7 @@ class HIPInducer {
   torch::Tensor init_node(const torch::Tensor& seeds) {
     device_ = seeds.device();
     count_ = 0;
-    nodes_so_far_ = torch::Tensor();
+    node_chunks_.clear();
     ensure_capacity(std::max<int64_t>(seeds.size(0), reserve_), true);
     return insert_and_assign(seeds);
   }
@@ -219,9 +219,9 @@ class HIPInducer {
       first_idx_ = torch::empty({capacity_}, opts);
       local_id_ = torch::empty({capacity_}, opts);
       reset_table();
-      if (!force_reset && nodes_so_far_.defined() && count_ > 0) {
+      if (!force_reset && count_ > 0) {
         // Re-insert the already-assigned unique nodes with their ids.
-        reinsert(nodes_so_far_);
+        reinsert(all_nodes());
       }
     } else if (force_reset) {
       reset_table();
@@ -269,14 +269,28 @@ class HIPInducer {
                          count_, uniq.data_ptr<int64_t>());
     }
     count_ += n_new;
-    nodes_so_far_ = nodes_so_far_.defined() && nodes_so_far_.size(0) > 0
-                        ? torch::cat({nodes_so_far_, uniq})
-                        : uniq;
+    if (n_new > 0) node_chunks_.push_back(uniq);
     return uniq;
   }
 
+  // Concatenate lazily: only the (rare) table-growth rebuild needs the full
+  // unique-node list, so per-hop O(n) device copies are avoided.
+  torch::Tensor all_nodes() {
+    if (node_chunks_.empty())
+      return torch::empty({0},
+                          torch::TensorOptions().dtype(torch::kInt64)
+                              .device(device_));
+    if (node_chunks_.size() > 1) {
+      auto merged = torch::cat(node_chunks_);
+      node_chunks_.clear();
+      node_chunks_.push_back(merged);
+    }
+    return node_chunks_[0];
+  }
+
   torch::Device device_{torch::kCPU};
-  torch::Tensor keys_, first_idx_, local_id_, nodes_so_far_;
+  torch::Tensor keys_, first_idx_, local_id_;
+  std::vector<torch::Tensor> node_chunks_;
   int64_t capacity_ = 0;
   int64_t count_ = 0;
   int64_t reserve_;

@@ -99,6 +99,10 @@ class DistNeighborSampler:
             self.sampler))
         self.event_loop = ConcurrentEventLoop(concurrency)
         self.event_loop.start_loop()
+        # every peer must have its callees registered before anyone samples
+        from .rpc import barrier
+
+        barrier()
 
     def shutdown(self):
         self.event_loop.shutdown_loop()
