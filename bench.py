@@ -41,6 +41,7 @@ def parse_args():
     p.add_argument("--graph-mode", type=str, default="CUDA",
                    choices=["CUDA", "ZERO_COPY", "CPU"])
     p.add_argument("--seed", type=int, default=42)
+    p.add_argument("--prefetch", type=int, default=3)
     return p.parse_args()
 
 
@@ -146,7 +147,8 @@ def main():
     seeds = torch.arange(args.nodes, device=device)
     loader = NeighborLoader(ds, fanout, input_nodes=seeds,
                             batch_size=args.batch_size, shuffle=True,
-                            device=device, to_device=device)
+                            device=device, to_device=device,
+                            prefetch=args.prefetch if has_gpu else 0)
     it = iter(loader)
 
     def one_step():

@@ -17,7 +17,7 @@ class NeighborLoader(NodeLoader):
                  device: Optional[torch.device] = None,
                  to_device: Optional[torch.device] = None,
                  edge_dir: Optional[str] = None, seed: Optional[int] = None,
-                 as_pyg_v1: bool = False, **kwargs):
+                 as_pyg_v1: bool = False, prefetch: int = 0, **kwargs):
         edge_dir = edge_dir or data.edge_dir
         sampler = NeighborSampler(
             data.get_graph() if not isinstance(data.graph, dict)
@@ -26,7 +26,7 @@ class NeighborLoader(NodeLoader):
             with_weight=with_weight, edge_dir=edge_dir, seed=seed)
         self.as_pyg_v1 = as_pyg_v1
         super().__init__(data, sampler, input_nodes, batch_size, shuffle,
-                         drop_last, with_edge, to_device)
+                         drop_last, with_edge, to_device, prefetch)
 
     def __next__(self):
         data = super().__next__()

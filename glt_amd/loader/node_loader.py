@@ -3,6 +3,7 @@
 Parity: reference python/loader/node_loader.py:54-115 (seed DataLoader +
 collate features/labels into Data/HeteroData).
 """
+import threading
 from typing import Optional, Union
 
 import torch
@@ -49,11 +50,84 @@ class _SeedIterator:
         return (self.n + self.batch_size - 1) // self.batch_size
 
 
+class _Prefetcher:
+    """Background sampling pipeline on a dedicated HIP stream.
+
+    The producer thread samples/collates batches on `side_stream` and
+    records an event per batch; the consumer makes its current stream wait
+    on that event, so sampling kernels for batch i+1 overlap the training
+    kernels of batch i on the GPU.
+    """
+
+    def __init__(self, loader, seed_iter, depth: int):
+        import queue as _q
+
+        self._queue: "_q.Queue" = _q.Queue(maxsize=depth)
+        self._loader = loader
+        self._seed_iter = seed_iter
+        self._stop = False
+        self._stream = torch.cuda.Stream()
+        self._thread = threading.Thread(target=self._run, daemon=True,
+                                        name="glt-prefetch")
+        self._thread.start()
+
+    def _run(self):
+        try:
+            with torch.cuda.stream(self._stream):
+                for seeds in self._seed_iter:
+                    if self._stop:
+                        return
+                    data = self._loader._produce(seeds)
+                    ev = torch.cuda.Event()
+                    ev.record(self._stream)
+                    self._queue.put((data, ev, None))
+            self._queue.put((None, None, None))
+        except BaseException as e:  # noqa: BLE001
+            self._queue.put((None, None, e))
+
+    def next(self):
+        data, ev, err = self._queue.get()
+        if err is not None:
+            raise err
+        if data is None:
+            raise StopIteration
+        torch.cuda.current_stream().wait_event(ev)
+        # allocations came from the side stream; register reuse on this one
+        cur = torch.cuda.current_stream()
+
+        def rec(t):
+            if t.is_cuda:
+                t.record_stream(cur)
+            return t
+
+        from ..pygcompat.data import _apply
+
+        for k, v in list(data.items()) if hasattr(data, "items") else []:
+            data[k] = _apply(v, rec)
+        return data
+
+    def stop(self):
+        self._stop = True
+        try:
+            while True:
+                self._queue.get_nowait()
+        except Exception:
+            pass
+
+
 class NodeLoader:
+    """Args beyond the reference surface:
+      prefetch: >0 enables pipelined sampling — a background thread samples
+        and collates the next `prefetch` batches on a dedicated side HIP
+        stream while the caller trains on the current batch (the
+        single-process analogue of the reference's subprocess sampling
+        pipeline)."""
+
     def __init__(self, data: Dataset, node_sampler: BaseSampler,
                  input_nodes, batch_size: int = 1, shuffle: bool = False,
                  drop_last: bool = False, with_edge: bool = False,
-                 to_device: Optional[torch.device] = None):
+                 to_device: Optional[torch.device] = None,
+                 prefetch: int = 0):
         self.data = data
         self.sampler = node_sampler
         self.input_nodes = NodeSamplerInput.cast(
@@ -65,23 +139,33 @@ class NodeLoader:
         self.drop_last = drop_last
         self.with_edge = with_edge
         self.to_device = to_device
+        self.prefetch = prefetch
+        self._prefetcher = None
 
     def __iter__(self):
         self._it = _SeedIterator(self.input_nodes.node, self.batch_size,
                                  self.shuffle, self.drop_last)
+        if self._prefetcher is not None:
+            self._prefetcher.stop()
+            self._prefetcher = None
+        if self.prefetch > 0 and torch.cuda.is_available():
+            self._prefetcher = _Prefetcher(self, self._it, self.prefetch)
         return self
 
     def __len__(self):
         return len(_SeedIterator(self.input_nodes.node, self.batch_size,
                                  False, self.drop_last))
 
-    def __next__(self):
-        seeds = next(self._it)
+    def _produce(self, seeds):
         inp = NodeSamplerInput(seeds, self.input_nodes.input_type)
         out = self.sampler.sample_from_nodes(inp)
         with trace_region("collate"):
-            result = self._collate_fn(out)
-        return result
+            return self._collate_fn(out)
+
+    def __next__(self):
+        if self._prefetcher is not None:
+            return self._prefetcher.next()
+        return self._produce(next(self._it))
 
     # -- feature/label collection ------------------------------------------
     def _collate_fn(self, out: Union[SamplerOutput, HeteroSamplerOutput]):
