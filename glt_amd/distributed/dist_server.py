@@ -34,6 +34,8 @@ class DistServer:
         self._producers: Dict[int, DistMpSamplingProducer] = {}
         self._channels: Dict[int, ShmChannel] = {}
         self._ends: Dict[int, int] = {}
+        self._epoch_done: Dict[int, bool] = {}
+        self._fetch_locks: Dict[int, threading.Lock] = {}
         self._next_id = 0
         self._lock = threading.Lock()
         self._exit = False
@@ -104,26 +106,38 @@ class DistServer:
         self._producers[pid] = producer
         self._channels[pid] = channel
         self._ends[pid] = 0
+        self._epoch_done[pid] = True  # nothing sampled until epoch start
+        self._fetch_locks[pid] = threading.Lock()
         return pid
 
     def start_new_epoch_sampling(self, producer_id: int):
-        self._ends[producer_id] = 0
-        self._producers[producer_id].produce_all()
+        with self._fetch_locks[producer_id]:
+            self._ends[producer_id] = 0
+            self._epoch_done[producer_id] = False
+            self._producers[producer_id].produce_all()
         return True
 
     def fetch_one_sampled_message(self, producer_id: int):
         """Returns the next SampleMessage, or {'#END'} once every worker of
-        this producer has finished the epoch."""
+        this producer has finished the epoch.  Serialized per producer so
+        concurrent client prefetches cannot race the END accounting, and
+        never blocks across an epoch boundary (a drained epoch returns END
+        immediately to every outstanding fetch)."""
         producer = self._producers[producer_id]
         channel = self._channels[producer_id]
-        while True:
-            msg = channel.recv(timeout_ms=300_000)
-            if END_KEY in msg:
-                self._ends[producer_id] += 1
-                if self._ends[producer_id] >= producer.num_expected_ends:
+        with self._fetch_locks[producer_id]:
+            while True:
+                if self._epoch_done[producer_id] and channel.empty():
                     return {END_KEY: torch.tensor([1])}
-                continue
-            return msg
+                msg = channel.recv(timeout_ms=300_000)
+                if END_KEY in msg:
+                    self._ends[producer_id] += 1
+                    if self._ends[producer_id] >= \
+                            producer.num_expected_ends:
+                        self._epoch_done[producer_id] = True
+                        return {END_KEY: torch.tensor([1])}
+                    continue
+                return msg
 
     def destroy_sampling_producer(self, producer_id: int):
         p = self._producers.pop(producer_id, None)
@@ -131,6 +145,8 @@ class DistServer:
             p.shutdown()
         self._channels.pop(producer_id, None)
         self._ends.pop(producer_id, None)
+        self._epoch_done.pop(producer_id, None)
+        self._fetch_locks.pop(producer_id, None)
         return True
 
 

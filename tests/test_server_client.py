@@ -1,0 +1,121 @@
+"""Server-client disaggregated mode: 1 server owning the dataset + 1 client
+pulling sampled batches over RPC (reference
+test/python/test_dist_neighbor_loader.py:213-291 capability)."""
+import multiprocessing as mp
+
+import pytest
+import torch
+
+VNUM = 40
+
+
+def _make_dataset():
+    import torch
+
+    from glt_amd.distributed import DistDataset
+    from glt_amd.data import Feature
+
+    rows, cols = [], []
+    for v in range(VNUM):
+        rows += [v, v]
+        cols += [(v + 1) % VNUM, (v + 2) % VNUM]
+    ds = DistDataset(num_partitions=1, partition_idx=0)
+    ds.init_graph(edge_index=torch.tensor([rows, cols]), graph_mode="CPU",
+                  num_nodes=VNUM)
+    feats = torch.arange(VNUM, dtype=torch.float32).unsqueeze(1).repeat(1, 8)
+    ds.node_features = Feature(feats, with_gpu=False)
+    ds.node_labels = torch.arange(VNUM)
+    from glt_amd.partition import GLTPartitionBook
+
+    ds.node_pb = GLTPartitionBook(torch.zeros(VNUM, dtype=torch.uint8))
+    return ds
+
+
+def _server_proc(port, q):
+    try:
+        from glt_amd.distributed import init_server, wait_and_shutdown_server
+
+        ds = _make_dataset()
+        init_server(num_servers=1, server_rank=0, dataset=ds,
+                    master_addr="127.0.0.1", master_port=port,
+                    num_clients=1)
+        wait_and_shutdown_server()
+        q.put(("server", None))
+    except Exception:  # noqa: BLE001
+        import traceback
+
+        q.put(("server", traceback.format_exc()))
+
+
+def _client_proc(port, q):
+    try:
+        import torch
+
+        from glt_amd.distributed import (DistNeighborLoader,
+                                         RemoteDistSamplingWorkerOptions,
+                                         init_client, request_server,
+                                         shutdown_client)
+
+        init_client(num_servers=1, num_clients=1, client_rank=0,
+                    master_addr="127.0.0.1", master_port=port)
+        # PyG remote-backend surface
+        meta = request_server(0, "get_dataset_meta")
+        assert meta["num_partitions"] == 1
+        feats = request_server(0, "get_node_feature",
+                               torch.tensor([3, 5]))
+        assert (feats == torch.tensor([[3.0] * 8, [5.0] * 8])).all()
+        labels = request_server(0, "get_node_label", torch.tensor([7]))
+        assert labels.tolist() == [7]
+        size = request_server(0, "get_tensor_size")
+        assert tuple(size) == (VNUM, 8)
+        pid = request_server(0, "get_node_partition_id",
+                             torch.tensor([1, 2]))
+        assert pid.tolist() == [0, 0]
+
+        # remote sampling loader
+        opts = RemoteDistSamplingWorkerOptions(
+            server_rank=0, num_workers=2, buffer_size="8MB",
+            buffer_capacity=16, prefetch_size=2)
+        loader = DistNeighborLoader(None, [2, 2],
+                                    input_nodes=torch.arange(VNUM),
+                                    batch_size=5,
+                                    worker_options=opts)
+        for epoch in range(2):
+            n = 0
+            for data in loader:
+                assert data.batch_size == 5
+                node = data.node
+                ei = data.edge_index
+                diff = (node[ei[1]] - node[ei[0]]) % VNUM
+                assert ((diff == 1) | (diff == 2)).all()
+                assert (data.x == node.float().unsqueeze(1)).all()
+                n += 1
+            assert n == 8, n
+        loader.shutdown()
+        shutdown_client()
+        q.put(("client", None))
+    except Exception:  # noqa: BLE001
+        import traceback
+
+        q.put(("client", traceback.format_exc()))
+
+
+@pytest.mark.timeout(300)
+def test_server_client_mode():
+    from glt_amd.utils import get_free_port
+
+    ctx = mp.get_context("spawn")
+    port = get_free_port()
+    q = ctx.Queue()
+    ps = ctx.Process(target=_server_proc, args=(port, q))
+    pc = ctx.Process(target=_client_proc, args=(port, q))
+    ps.start()
+    pc.start()
+    results = [q.get(timeout=280) for _ in range(2)]
+    pc.join(timeout=30)
+    ps.join(timeout=30)
+    for p in (ps, pc):
+        if p.is_alive():
+            p.terminate()
+    for who, err in results:
+        assert err is None, f"{who}:\n{err}"
