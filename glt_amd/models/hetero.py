@@ -57,9 +57,10 @@ def conv_bipartite(conv, x_tgt, x_src, edge_index):
 class RGNN(nn.Module):
     """Relational GNN over hetero batches; model='rgat' or 'rsage'.
 
-    Per layer: per-edge-type conv (GAT or SAGE) combined per node type by
-    summation, followed by ReLU/dropout; a final linear head on the
-    predicted node type.
+    Per layer: per-edge-type conv (GAT or SAGE) summed per target node
+    type, plus a per-node-type self projection so every type advances to
+    the layer's output dim even when it receives no messages; ReLU/dropout
+    between layers; linear head on the predicted node type.
     """
 
     def __init__(self, etypes: List[EdgeType], in_dim: int, h_dim: int,
@@ -68,8 +69,11 @@ class RGNN(nn.Module):
                  node_types: Optional[List[NodeType]] = None):
         super().__init__()
         self.model = model
+        if node_types is None:
+            node_types = sorted({t for et in etypes for t in (et[0], et[2])})
+        self.node_types = node_types
         self.layers = nn.ModuleList()
-        self.lin_in = None
+        self.self_lins = nn.ModuleList()
         dims = [in_dim] + [h_dim] * num_layers
         for li in range(num_layers):
             convs = {}
@@ -78,8 +82,12 @@ class RGNN(nn.Module):
                     convs[et] = GATConv(dims[li], dims[li + 1] // n_heads,
                                         heads=n_heads, concat=True)
                 else:
-                    convs[et] = SAGEConv(dims[li], dims[li + 1])
+                    convs[et] = SAGEConv(dims[li], dims[li + 1],
+                                         root_weight=False)
             self.layers.append(HeteroConv(convs))
+            self.self_lins.append(nn.ModuleDict(
+                {t: nn.Linear(dims[li], dims[li + 1])
+                 for t in node_types}))
         self.head = nn.Linear(h_dim, out_dim)
         self.dropout = dropout
 
@@ -88,12 +96,21 @@ class RGNN(nn.Module):
                 predict_type: Optional[NodeType] = None):
         h = x_dict
         for i, layer in enumerate(self.layers):
-            h_new = layer(h, edge_index_dict)
-            # keep untouched types' features flowing
-            h = {**h, **{t: F.relu(v) for t, v in h_new.items()}}
-            h = {t: F.dropout(v, p=self.dropout, training=self.training)
-                 for t, v in h.items()}
+            h_conv = layer(h, edge_index_dict)
+            h_next = {}
+            for t, v in h.items():
+                if t not in self.self_lins[i]:
+                    continue
+                out = self.self_lins[i][t](v)
+                hc = h_conv.get(t)
+                if hc is not None:
+                    # conv output may cover fewer rows than the projection
+                    # (message targets only); add on the common prefix
+                    m = min(hc.size(0), out.size(0))
+                    out = torch.cat([out[:m] + hc[:m], out[m:]], dim=0)
+                h_next[t] = F.dropout(F.relu(out), p=self.dropout,
+                                      training=self.training)
+            h = h_next
         if predict_type is not None:
             return self.head(h[predict_type])
-        return {t: self.head(v) if v.size(-1) == self.head.in_features else v
-                for t, v in h.items()}
+        return {t: self.head(v) for t, v in h.items()}
