@@ -188,6 +188,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("rows", &FeatureStorePy::rows)
       .def("dim", &FeatureStorePy::dim);
 
+  m.def("segment_mean_fwd", &hip_segment_mean_fwd);
+  m.def("segment_mean_bwd", &hip_segment_mean_bwd);
+
   // Memory plumbing
   m.def("host_mapped_view", &host_mapped_view, py::arg("src"),
         py::arg("device") = 0);

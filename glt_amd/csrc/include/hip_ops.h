@@ -60,6 +60,16 @@ hip_stitch_sample_results(int64_t ids_count,
                           const std::vector<torch::Tensor>& nbrs_num_list,
                           const std::vector<torch::Tensor>& eids_list);
 
+// --- fused segment aggregation (hip_segment.hip) ----------------------------
+torch::Tensor hip_segment_mean_fwd(const torch::Tensor& x,
+                                   const torch::Tensor& col,
+                                   const torch::Tensor& offsets,
+                                   int64_t n_tgt);
+torch::Tensor hip_segment_mean_bwd(const torch::Tensor& dy,
+                                   const torch::Tensor& col,
+                                   const torch::Tensor& offsets,
+                                   int64_t n_src);
+
 // --- memory plumbing (hip_mem.hip) -----------------------------------------
 // Device-dtype alias of (pinned/registered) host memory; keeps `src` alive.
 torch::Tensor host_mapped_view(const torch::Tensor& src, int64_t device_index);

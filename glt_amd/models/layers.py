@@ -33,12 +33,20 @@ class SAGEConv(nn.Module):
                       if root_weight else None)
 
     def forward(self, x: torch.Tensor, edge_index: torch.Tensor,
-                num_target: int = None) -> torch.Tensor:
+                num_target: int = None, sorted_by_target: bool = True
+                ) -> torch.Tensor:
         n = num_target if num_target is not None else x.size(0)
         tgt, src = edge_index[0], edge_index[1]
-        agg = x.new_zeros(n, x.size(1))
-        agg.index_add_(0, tgt, x.index_select(0, src))
-        agg = agg / _degree(tgt, n).unsqueeze(1).to(x.dtype)
+        if (x.is_cuda and x.dtype == torch.float32 and sorted_by_target):
+            # fused wave-per-row segment mean (glt_amd batches are sorted
+            # by target local id by construction)
+            from ..ops import segment_mean
+
+            agg = segment_mean(x, tgt, src, n)
+        else:
+            agg = x.new_zeros(n, x.size(1))
+            agg.index_add_(0, tgt, x.index_select(0, src))
+            agg = agg / _degree(tgt, n).unsqueeze(1).to(x.dtype)
         out = self.lin_l(agg)
         if self.lin_r is not None:
             out = out + self.lin_r(x[:n])

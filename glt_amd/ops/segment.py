@@ -1,0 +1,38 @@
+"""Autograd wrapper over the fused HIP segment-mean kernel.
+
+Used by SAGEConv when the batch lives on the GPU and edges are sorted by
+target (the glt_amd sampler's native edge order); falls back to torch
+index_add on CPU or unsorted input.
+"""
+import torch
+
+
+class _SegmentMean(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, col, offsets, n_src):
+        from .. import _C
+
+        ctx.save_for_backward(col, offsets)
+        ctx.n_src = n_src
+        return _C.segment_mean_fwd(x, col, offsets, offsets.numel() - 1)
+
+    @staticmethod
+    def backward(ctx, dy):
+        from .. import _C
+
+        col, offsets = ctx.saved_tensors
+        dx = _C.segment_mean_bwd(dy.contiguous(), col, offsets, ctx.n_src)
+        return dx, None, None, None
+
+
+def segment_mean(x: torch.Tensor, tgt: torch.Tensor, src: torch.Tensor,
+                 n_tgt: int) -> torch.Tensor:
+    """Mean of x[src[e]] over contiguous tgt segments.
+
+    Requires tgt ascending (glt_amd batches satisfy this); returns
+    [n_tgt, F].
+    """
+    boundaries = torch.arange(n_tgt + 1, device=tgt.device)
+    offsets = torch.searchsorted(tgt, boundaries)
+    return _SegmentMean.apply(x.contiguous(), src.contiguous(), offsets,
+                              x.size(0))
