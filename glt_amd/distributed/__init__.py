@@ -10,6 +10,7 @@ from .dist_neighbor_sampler import DistNeighborSampler
 from .dist_options import (CollocatedDistSamplingWorkerOptions,
                            MpDistSamplingWorkerOptions,
                            RemoteDistSamplingWorkerOptions)
+from .dist_random_partitioner import DistRandomPartitioner
 from .dist_sampling_producer import (DistCollocatedSamplingProducer,
                                      DistMpSamplingProducer)
 from .dist_server import (DistServer, get_server, init_server,
@@ -25,7 +26,7 @@ __all__ = [
     "DistContext", "DistRole", "assign_server_by_order", "get_context",
     "init_worker_group", "DistDataset", "DistFeature", "DistGraph",
     "DistLoader", "DistLinkNeighborLoader", "DistNeighborLoader",
-    "DistSubGraphLoader", "DistNeighborSampler",
+    "DistSubGraphLoader", "DistNeighborSampler", "DistRandomPartitioner",
     "CollocatedDistSamplingWorkerOptions", "MpDistSamplingWorkerOptions",
     "RemoteDistSamplingWorkerOptions", "DistCollocatedSamplingProducer",
     "DistMpSamplingProducer", "DistServer", "get_server", "init_server",

@@ -261,3 +261,64 @@ def test_frequency_partitioner(tmp_path, ring_graph):
     # cache rows hold the features of the cached ids
     assert (node_feat.cache_feats ==
             node_feat.cache_ids.float().unsqueeze(1)).all()
+
+
+def _worker_dist_partitioner(rank, world, port, q):
+    try:
+        import torch
+
+        import glt_amd
+        from glt_amd.distributed import init_worker_group
+        from glt_amd.distributed.rpc import init_rpc, shutdown_rpc
+        from glt_amd.distributed.dist_random_partitioner import \
+            DistRandomPartitioner
+
+        init_worker_group(world, rank)
+        init_rpc("127.0.0.1", port)
+        n = 100
+        # rank r holds edges with src in [r*50, r*50+50)
+        src = torch.arange(rank * 50, rank * 50 + 50).repeat_interleave(2)
+        dst = (src + torch.randint(1, 50, (100,))) % n
+        feats = torch.arange(rank * 50, rank * 50 + 50,
+                             dtype=torch.float32).unsqueeze(1)
+        ids = torch.arange(rank * 50, rank * 50 + 50)
+        eids = torch.arange(rank * 100, rank * 100 + 100)
+        p = DistRandomPartitioner(n, torch.stack([src, dst]),
+                                  local_eids=eids, local_node_feat=feats,
+                                  local_node_ids=ids, seed=5)
+        pb, graph, feat = p.partition()
+        # every received edge belongs here (by src)
+        assert (pb[graph.edge_index[0]] == rank).all()
+        # features closed-form + owned here
+        assert (pb[feat.ids] == rank).all()
+        assert (feat.feats == feat.ids.float().unsqueeze(1)).all()
+        total_edges = torch.tensor([graph.edge_index.size(1)])
+        from glt_amd.distributed import barrier
+
+        barrier()
+        shutdown_rpc()
+        q.put((rank, None))
+    except Exception:  # noqa: BLE001
+        import traceback
+
+        q.put((rank, traceback.format_exc()))
+
+
+@pytest.mark.timeout(180)
+def test_dist_random_partitioner():
+    from glt_amd.utils import get_free_port
+
+    ctx = mp.get_context("spawn")
+    port = get_free_port()
+    q = ctx.Queue()
+    ps = [ctx.Process(target=_worker_dist_partitioner,
+                      args=(r, 2, port, q)) for r in range(2)]
+    for p in ps:
+        p.start()
+    results = [q.get(timeout=150) for _ in range(2)]
+    for p in ps:
+        p.join(timeout=30)
+        if p.is_alive():
+            p.terminate()
+    for rank, err in results:
+        assert err is None, f"rank {rank}:\n{err}"
