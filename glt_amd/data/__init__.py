@@ -3,6 +3,9 @@ from .feature import Feature, DeviceGroup
 from .dataset import Dataset
 from .reorder import sort_by_in_degree
 from .unified_tensor import UnifiedTensor
+from .table_dataset import TableDataset
+from . import vineyard_utils
 
 __all__ = ["Topology", "Graph", "Feature", "DeviceGroup", "Dataset",
-           "sort_by_in_degree", "UnifiedTensor"]
+           "sort_by_in_degree", "UnifiedTensor", "TableDataset",
+           "vineyard_utils"]

@@ -136,6 +136,29 @@ def save_feature_partition(output_dir, partition_idx,
         torch.save(part.cache_ids, os.path.join(sub, "cache_ids.pt"))
 
 
+def save_graph_cache(output_dir, graph_partition_list, etype=None,
+                     with_edge_feat: bool = False):
+    """Whole-topology cache for `graph_caching=True` datasets (parity:
+    reference partition/base.py:93-118): concatenates every partition's COO
+    into one root-level graph/ directory."""
+    if not graph_partition_list:
+        return
+    sub = os.path.join(output_dir, "graph")
+    if etype is not None:
+        sub = os.path.join(sub, as_str(etype))
+    ensure_dir(sub)
+    rows = torch.cat([g.edge_index[0] for g in graph_partition_list])
+    cols = torch.cat([g.edge_index[1] for g in graph_partition_list])
+    torch.save(rows, os.path.join(sub, "rows.pt"))
+    torch.save(cols, os.path.join(sub, "cols.pt"))
+    if with_edge_feat:
+        torch.save(torch.cat([g.eids for g in graph_partition_list]),
+                   os.path.join(sub, "eids.pt"))
+    if graph_partition_list[0].weights is not None:
+        torch.save(torch.cat([g.weights for g in graph_partition_list]),
+                   os.path.join(sub, "weights.pt"))
+
+
 # ---------------------------------------------------------------------------
 # partitioner
 # ---------------------------------------------------------------------------

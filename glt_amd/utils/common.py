@@ -29,6 +29,12 @@ def seed_everything(seed: int):
         pass
 
 
+def tensor_equal_with_device(a: torch.Tensor, b: torch.Tensor) -> bool:
+    """Equality including device placement (reference glt.utils helper)."""
+    return a.device == b.device and a.shape == b.shape and bool(
+        (a == b).all())
+
+
 def id2idx(ids: Union[torch.Tensor, List[int]]) -> torch.Tensor:
     """Dense global-id -> position map (parity: utils/tensor.py:30-39)."""
     if not torch.is_tensor(ids):
