@@ -159,7 +159,7 @@ class HIPInducer {
     auto nodes = insert_and_assign(nbrs);
     // offsets for row expansion
     auto offsets = torch::zeros({ns + 1}, srcs.options());
-    offsets.narrow(0, 1, ns).copy_(torch::cumsum(nbrs_num, 0));
+    { auto v = offsets.narrow(0, 1, ns); torch::cumsum_out(v, nbrs_num, 0); }
     auto rows = torch::empty({ne}, srcs.options());
     auto cols = torch::empty({ne}, srcs.options());
     if (ne > 0) {
@@ -375,7 +375,7 @@ hip_node_subgraph(const torch::Tensor& indptr, const torch::Tensor& indices,
   auto stream = current_stream();
   auto degs = hip_lookup_degree(indptr, uniq);  // declared in hip_ops.h
   auto edge_offsets = torch::zeros({n + 1}, nodes.options());
-  if (n > 0) edge_offsets.narrow(0, 1, n).copy_(torch::cumsum(degs, 0));
+  if (n > 0) { auto v = edge_offsets.narrow(0, 1, n); torch::cumsum_out(v, degs, 0); }
   const int64_t total = n > 0 ? edge_offsets[n].item<int64_t>() : 0;
   auto flags = torch::zeros({std::max<int64_t>(total, 1)}, nodes.options());
   auto cols_tmp = torch::empty({std::max<int64_t>(total, 1)}, nodes.options());
@@ -447,7 +447,7 @@ hip_stitch_sample_results(int64_t ids_count,
   for (size_t p = 0; p < P; ++p)
     nbrs_num.index_copy_(0, idx_list[p], nbrs_num_list[p]);
   auto out_offsets = torch::zeros({ids_count + 1}, opts);
-  out_offsets.narrow(0, 1, ids_count).copy_(torch::cumsum(nbrs_num, 0));
+  { auto v = out_offsets.narrow(0, 1, ids_count); torch::cumsum_out(v, nbrs_num, 0); }
   const int64_t total = out_offsets[ids_count].item<int64_t>();
   auto nbrs = torch::zeros({total}, opts);
   auto eids = with_edge ? torch::zeros({total}, opts) : torch::Tensor();
@@ -456,7 +456,7 @@ hip_stitch_sample_results(int64_t ids_count,
     const int64_t m = idx_list[p].size(0);
     if (m == 0) continue;
     auto p_offsets = torch::zeros({m + 1}, opts);
-    p_offsets.narrow(0, 1, m).copy_(torch::cumsum(nbrs_num_list[p], 0));
+    { auto v = p_offsets.narrow(0, 1, m); torch::cumsum_out(v, nbrs_num_list[p], 0); }
     const int64_t p_total = nbrs_list[p].size(0);
     if (p_total == 0) continue;
     hipLaunchKernelGGL(stitch_kernel, dim3(grid_for(p_total)), dim3(kBlock),

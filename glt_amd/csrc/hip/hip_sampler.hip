@@ -232,7 +232,7 @@ hip_sample_neighbors(const torch::Tensor& indptr, const torch::Tensor& indices,
                        counts.data_ptr<int64_t>());
   }
   auto offsets = torch::zeros({bs + 1}, seeds.options());
-  offsets.narrow(0, 1, bs).copy_(torch::cumsum(counts, 0));
+  { auto v = offsets.narrow(0, 1, bs); torch::cumsum_out(v, counts, 0); }
   const int64_t total = bs > 0 ? offsets[bs].item<int64_t>() : 0;  // hop sync
 
   auto nbrs = torch::empty({total}, seeds.options());
@@ -375,7 +375,7 @@ torch::Tensor hip_cal_nbr_prob(const torch::Tensor& indptr,
   const int64_t num_rows = indptr.size(0) - 1;
   auto degs = hip_lookup_degree(indptr, nodes);
   auto edge_offsets = torch::zeros({n + 1}, nodes.options());
-  edge_offsets.narrow(0, 1, n).copy_(torch::cumsum(degs, 0));
+  { auto v = edge_offsets.narrow(0, 1, n); torch::cumsum_out(v, degs, 0); }
   const int64_t total = n > 0 ? edge_offsets[n].item<int64_t>() : 0;
   auto log_acc = torch::zeros_like(last_prob);
   if (total > 0) {

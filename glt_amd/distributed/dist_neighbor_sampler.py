@@ -195,7 +195,7 @@ class DistNeighborSampler:
     # ------------------------------------------------------------------
     async def _multihop(self, seeds: torch.Tensor, metadata=None
                         ) -> SamplerOutput:
-        inducer = self.sampler._make_inducer()
+        inducer = self.sampler._acquire_inducer()
         uniq = inducer.init_node(seeds.to(self.sampler.device))
         out_nodes, num_nodes, num_edges = [uniq], [uniq.numel()], []
         rows, cols, eids = [], [], []
@@ -212,6 +212,7 @@ class DistNeighborSampler:
                 eids.append(out.edge)
             srcs = nodes
         dev = self.sampler.device
+        self.sampler._release_inducer(inducer)
         return SamplerOutput(
             node=torch.cat(out_nodes),
             row=torch.cat(rows) if rows else torch.empty(

@@ -96,6 +96,7 @@ class NeighborSampler(BaseSampler):
         self.edge_dir = edge_dir
         self.is_hetero = isinstance(graph, dict)
         self._lock = threading.Lock()
+        self._inducer_pool = []
         if seed is not None:
             _C.manual_seed(seed)
         g0 = next(iter(graph.values())) if self.is_hetero else graph
@@ -111,7 +112,21 @@ class NeighborSampler(BaseSampler):
     def _make_inducer(self):
         if self._cpu_mode:
             return self._C.CPUInducer(1024)
-        return self._C.DeviceInducer(4096)
+        return self._C.DeviceInducer(65536)
+
+    def _acquire_inducer(self):
+        """Reuse inducers across batches: their hash-table capacity is
+        retained, so steady-state batches skip allocation and the
+        growth-rebuild inserts."""
+        with self._lock:
+            if self._inducer_pool:
+                return self._inducer_pool.pop()
+        return self._make_inducer()
+
+    def _release_inducer(self, ind):
+        with self._lock:
+            if len(self._inducer_pool) < 8:
+                self._inducer_pool.append(ind)
 
     def _seeds_to_device(self, seeds: torch.Tensor) -> torch.Tensor:
         return seeds.long().to(self._sample_device, non_blocking=True)
@@ -141,7 +156,7 @@ class NeighborSampler(BaseSampler):
 
     def _sample_from_nodes(self, seeds: torch.Tensor,
                            metadata=None) -> SamplerOutput:
-        inducer = self._make_inducer()
+        inducer = self._acquire_inducer()
         uniq_seeds = inducer.init_node(seeds)
         out_nodes = [uniq_seeds]
         num_nodes = [uniq_seeds.numel()]
@@ -165,6 +180,7 @@ class NeighborSampler(BaseSampler):
         col = torch.cat(cols) if cols else torch.empty(
             0, dtype=torch.long, device=self.device)
         edge = torch.cat(eids) if eids else None
+        self._release_inducer(inducer)
         return SamplerOutput(
             node=node, row=row, col=col, edge=edge, batch=uniq_seeds,
             num_sampled_nodes=num_nodes, num_sampled_edges=num_edges,
