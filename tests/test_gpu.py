@@ -271,3 +271,29 @@ def test_sageconv_fused_matches_fallback():
     fused = conv(x, ei, num_target=50)
     fallback = conv(x, ei, num_target=50, sorted_by_target=False)
     assert torch.allclose(fused, fallback, atol=1e-4)
+
+
+def test_weighted_sampler_gpu():
+    glt_amd.seed_everything(3)
+    topo = Topology(torch.tensor([[0, 0], [1, 2]]),
+                    edge_weights=torch.tensor([9.0, 1.0]), num_nodes=3)
+    indptr = topo.indptr.cuda()
+    indices = topo.indices.cuda()
+    w = topo.edge_weights.cuda()
+    seeds = torch.zeros(4000, dtype=torch.long, device="cuda")
+    nbrs, num, _ = _C.sample_neighbors(indptr, indices, seeds, 1,
+                                       edge_weights=w, weighted=True)
+    frac = (nbrs == 1).float().mean().item()
+    assert 0.82 < frac < 0.97, frac
+
+
+def test_sample_prob_gpu(ring_graph):
+    from glt_amd.data import Graph
+    from glt_amd.sampler import NeighborSampler
+
+    topo = Topology(ring_graph["edge_index"], num_nodes=40)
+    g = Graph(topo, mode="CUDA", device=0)
+    s = NeighborSampler(g, [1])
+    prob = s.sample_prob(torch.tensor([0], device="cuda"), 40)
+    assert abs(prob[1].item() - 0.5) < 1e-4
+    assert prob[0].item() == 1.0
