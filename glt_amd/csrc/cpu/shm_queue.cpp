@@ -162,6 +162,12 @@ ShmQueue::Ticket ShmQueue::reserve(uint64_t size) {
       uint64_t v = meta_->alloc_off;
       const uint64_t phys = v % meta_->ring_bytes;
       if (phys + size > meta_->ring_bytes) v += meta_->ring_bytes - phys;  // skip tail fragment
+      if (meta_->head_desc == meta_->tail_desc) {
+        // queue empty: the skipped tail fragment holds no live data, so the
+        // free boundary may jump with the allocation pointer (otherwise a
+        // message with size > phys of an empty ring deadlocks).
+        meta_->free_off = v;
+      }
       if (v + size - meta_->free_off <= meta_->ring_bytes) {
         const uint32_t slot = (uint32_t)(meta_->head_desc % meta_->capacity);
         Desc& d = descs_[slot];
