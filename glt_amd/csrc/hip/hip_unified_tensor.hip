@@ -30,37 +30,37 @@ struct SegTable {
   int n;
 };
 
-// One wave per output row; VEC = bytes per lane per step (16/8/4/1).
+// Element-level gather: thread t copies VEC bytes of slot (t % slots) of
+// output row (t / slots).  Full lane utilization for short feature rows
+// (a 400-byte fp32 dim-100 row is only 25 16-byte slots — a wave-per-row
+// tiling would idle 39 of 64 lanes); writes stay perfectly coalesced and
+// row reads are contiguous VEC-chunks resolved per segment.
 template <int VEC>
 __global__ void gather_rows_kernel(SegTable segs, int64_t row_bytes,
                                    const int64_t* __restrict__ rows,
                                    int64_t n, char* __restrict__ out) {
-  const int lane = threadIdx.x & (kWave - 1);
-  const int64_t wave =
-      (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) / kWave;
-  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) / kWave;
-  for (int64_t i = wave; i < n; i += n_waves) {
+  const int64_t slots = row_bytes / VEC;
+  const int64_t total = n * slots;
+  for (int64_t idx = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       idx < total; idx += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t i = idx / slots;
+    const int64_t j = idx - i * slots;
     const int64_t r = rows[i];
-    // Resolve owning segment (few segments: linear scan, wave-uniform).
     int s = 0;
     while (s + 1 < segs.n && r >= segs.row_start[s + 1]) ++s;
-    const char* src =
-        segs.base[s] + (r - segs.row_start[s]) * row_bytes;
+    const char* src = segs.base[s] + (r - segs.row_start[s]) * row_bytes;
     char* dst = out + i * row_bytes;
-    const int64_t steps = row_bytes / VEC;
-    for (int64_t j = lane; j < steps; j += kWave) {
-      if (VEC == 16) {
-        reinterpret_cast<uint4*>(dst)[j] =
-            reinterpret_cast<const uint4*>(src)[j];
-      } else if (VEC == 8) {
-        reinterpret_cast<uint2*>(dst)[j] =
-            reinterpret_cast<const uint2*>(src)[j];
-      } else if (VEC == 4) {
-        reinterpret_cast<uint32_t*>(dst)[j] =
-            reinterpret_cast<const uint32_t*>(src)[j];
-      } else {
-        dst[j] = src[j];
-      }
+    if (VEC == 16) {
+      reinterpret_cast<uint4*>(dst)[j] =
+          reinterpret_cast<const uint4*>(src)[j];
+    } else if (VEC == 8) {
+      reinterpret_cast<uint2*>(dst)[j] =
+          reinterpret_cast<const uint2*>(src)[j];
+    } else if (VEC == 4) {
+      reinterpret_cast<uint32_t*>(dst)[j] =
+          reinterpret_cast<const uint32_t*>(src)[j];
+    } else {
+      dst[j] = src[j];
     }
   }
 }
@@ -109,11 +109,10 @@ class UnifiedFeatureStore {
     t.row_start[t.n] = acc;
 
     const int64_t row_bytes = dim_ * (int64_t)elementSize(dtype_);
-    // Enough waves to fill 256 CUs several times over.
-    const int64_t waves = std::min<int64_t>(n, (int64_t)kMaxBlocks * 4);
-    const int blocks =
-        (int)std::min<int64_t>((waves * kWave + kBlock - 1) / kBlock,
-                               kMaxBlocks);
+    const int64_t vec = row_bytes % 16 == 0 ? 16
+                        : row_bytes % 8 == 0 ? 8
+                        : row_bytes % 4 == 0 ? 4 : 1;
+    const int blocks = grid_for(n * (row_bytes / vec));
     auto stream = current_stream();
     char* out_p = reinterpret_cast<char*>(out.data_ptr());
     const int64_t* rows_p = rows.data_ptr<int64_t>();
