@@ -322,3 +322,58 @@ def test_dist_random_partitioner():
             p.terminate()
     for rank, err in results:
         assert err is None, f"rank {rank}:\n{err}"
+
+
+def _worker_all2all(rank, world, port, q):
+    try:
+        import os
+
+        import torch
+        import torch.distributed as dist
+
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        from glt_amd.data import Feature
+        from glt_amd.distributed.dist_feature import DistFeature
+        from glt_amd.partition import GLTPartitionBook
+
+        n = 40
+        node_pb = GLTPartitionBook(torch.arange(n) % 2)
+        feats = torch.arange(n, dtype=torch.float32).unsqueeze(1).repeat(1, 8)
+        local_ids = torch.arange(rank, n, 2)
+        id2index = torch.full((n,), -1, dtype=torch.long)
+        id2index[local_ids] = torch.arange(local_ids.numel())
+        df = DistFeature(2, rank, Feature(feats[local_ids], with_gpu=False,
+                                          id2index=id2index),
+                         None, node_pb, None)
+        torch.manual_seed(rank)
+        ids = torch.randint(0, n, (33,))
+        out = df.all2all_get("node", ids)
+        assert torch.equal(out, feats[ids]), (out, feats[ids])
+        dist.destroy_process_group()
+        q.put((rank, None))
+    except Exception:  # noqa: BLE001
+        import traceback
+
+        q.put((rank, traceback.format_exc()))
+
+
+@pytest.mark.timeout(120)
+def test_dist_feature_all2all():
+    from glt_amd.utils import get_free_port
+
+    ctx = mp.get_context("spawn")
+    port = get_free_port()
+    q = ctx.Queue()
+    ps = [ctx.Process(target=_worker_all2all, args=(r, 2, port, q))
+          for r in range(2)]
+    for p in ps:
+        p.start()
+    results = [q.get(timeout=100) for _ in range(2)]
+    for p in ps:
+        p.join(timeout=20)
+        if p.is_alive():
+            p.terminate()
+    for rank, err in results:
+        assert err is None, f"rank {rank}:\n{err}"
