@@ -44,6 +44,9 @@ def parse_args():
                    choices=["CUDA", "ZERO_COPY", "CPU"])
     p.add_argument("--seed", type=int, default=42)
     p.add_argument("--prefetch", type=int, default=3)
+    p.add_argument("--amp", action="store_true",
+                   help="bf16 autocast for model math (default fp32, "
+                        "matching the reference)")
     return p.parse_args()
 
 
@@ -153,6 +156,13 @@ def main():
                             prefetch=args.prefetch if has_gpu else 0)
     it = iter(loader)
 
+    import contextlib
+
+    def amp_ctx():
+        if args.amp and has_gpu:
+            return torch.autocast("cuda", dtype=torch.bfloat16)
+        return contextlib.nullcontext()
+
     def one_step():
         nonlocal it
         try:
@@ -161,9 +171,10 @@ def main():
             it = iter(loader)
             data = next(it)
         opt.zero_grad(set_to_none=True)
-        out = model(data.x, data.edge_index, data.num_sampled_nodes,
-                    data.num_sampled_edges)[:data.batch_size]
-        loss = F.cross_entropy(out, data.y[:data.batch_size])
+        with amp_ctx():
+            out = model(data.x, data.edge_index, data.num_sampled_nodes,
+                        data.num_sampled_edges)[:data.batch_size]
+            loss = F.cross_entropy(out, data.y[:data.batch_size])
         loss.backward()
         opt.step()
         return loss
@@ -203,7 +214,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "fp32",
+            "dtype": "bf16" if args.amp else "fp32",
             "data": "synthetic",
             "config": {
                 "model": "GraphSAGE(3x256)",

@@ -36,8 +36,10 @@ class HeteroConv(nn.Module):
             if x_tgt is None or x_src is None:
                 continue
             conv = self.convs[key]
-            h = conv((x_tgt, x_src), ei) if hasattr(conv, "bipartite") \
-                else conv_bipartite(conv, x_tgt, x_src, ei)
+            try:
+                h = conv((x_tgt, x_src), ei)
+            except TypeError:
+                h = conv_bipartite(conv, x_tgt, x_src, ei)
             out.setdefault(tgt_t, []).append(h)
         result = {}
         for t, hs in out.items():

@@ -32,9 +32,27 @@ class SAGEConv(nn.Module):
         self.lin_r = (nn.Linear(in_channels, out_channels, bias=False)
                       if root_weight else None)
 
-    def forward(self, x: torch.Tensor, edge_index: torch.Tensor,
+    def forward(self, x, edge_index: torch.Tensor,
                 num_target: int = None, sorted_by_target: bool = True
                 ) -> torch.Tensor:
+        """x: [n, F] or (x_target, x_source) for bipartite relations."""
+        if isinstance(x, tuple):
+            x_tgt, x_src = x
+            n = num_target if num_target is not None else x_tgt.size(0)
+            tgt, src = edge_index[0], edge_index[1]
+            if x_src.is_cuda and x_src.dtype == torch.float32 and \
+                    sorted_by_target:
+                from ..ops import segment_mean
+
+                agg = segment_mean(x_src, tgt, src, n)
+            else:
+                agg = x_src.new_zeros(n, x_src.size(1))
+                agg.index_add_(0, tgt, x_src.index_select(0, src))
+                agg = agg / _degree(tgt, n).unsqueeze(1).to(x_src.dtype)
+            out = self.lin_l(agg)
+            if self.lin_r is not None:
+                out = out + self.lin_r(x_tgt[:n])
+            return out
         n = num_target if num_target is not None else x.size(0)
         tgt, src = edge_index[0], edge_index[1]
         if (x.is_cuda and x.dtype == torch.float32 and sorted_by_target):
@@ -97,12 +115,24 @@ class GATConv(nn.Module):
         nn.init.xavier_uniform_(self.att_dst)
 
     def forward(self, x, edge_index, num_target: int = None):
-        n = x.size(0)
-        nt = num_target if num_target is not None else n
-        h = self.lin(x).view(n, self.heads, self.out_channels)
+        """x: [n, F] or a (x_target, x_source) tuple for bipartite edge
+        sets (hetero relations); edge_index[0] indexes the target side,
+        edge_index[1] the source side."""
+        if isinstance(x, tuple):
+            x_tgt, x_src = x
+            nt = num_target if num_target is not None else x_tgt.size(0)
+            h_tgt = self.lin(x_tgt).view(x_tgt.size(0), self.heads,
+                                         self.out_channels)
+            h_src = self.lin(x_src).view(x_src.size(0), self.heads,
+                                         self.out_channels)
+        else:
+            nt = num_target if num_target is not None else x.size(0)
+            h_tgt = h_src = self.lin(x).view(x.size(0), self.heads,
+                                             self.out_channels)
         tgt, src = edge_index[0], edge_index[1]
-        alpha_src = (h * self.att_src).sum(-1)  # [n, H]
-        alpha_dst = (h * self.att_dst).sum(-1)
+        h = h_src
+        alpha_src = (h_src * self.att_src).sum(-1)
+        alpha_dst = (h_tgt * self.att_dst).sum(-1)
         e = alpha_dst[tgt] + alpha_src[src]  # [E, H]
         e = F.leaky_relu(e, self.negative_slope)
         # scatter softmax over tgt
