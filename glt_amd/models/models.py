@@ -76,10 +76,18 @@ class GAT(nn.Module):
         self.convs.append(GATConv(dim, out_channels, heads=1, concat=False))
         self.dropout = dropout
 
-    def forward(self, x, edge_index):
+    def forward(self, x, edge_index, num_sampled_nodes=None,
+                num_sampled_edges=None):
+        L = len(self.convs)
         for i, conv in enumerate(self.convs):
-            x = conv(x, edge_index)
-            if i < len(self.convs) - 1:
+            trim = _layer_trim(num_sampled_nodes, num_sampled_edges, L, i)
+            if trim is not None:
+                n_in, n_edges, n_out = trim
+                x = conv(x[:n_in], edge_index[:, :n_edges],
+                         num_target=n_out)
+            else:
+                x = conv(x, edge_index)
+            if i < L - 1:
                 x = F.elu(x)
                 x = F.dropout(x, p=self.dropout, training=self.training)
         return x
@@ -98,10 +106,18 @@ class GCN(nn.Module):
             self.convs.append(GCNConv(dims[i], dims[i + 1]))
         self.dropout = dropout
 
-    def forward(self, x, edge_index):
+    def forward(self, x, edge_index, num_sampled_nodes=None,
+                num_sampled_edges=None):
+        L = len(self.convs)
         for i, conv in enumerate(self.convs):
-            x = conv(x, edge_index)
-            if i < len(self.convs) - 1:
+            trim = _layer_trim(num_sampled_nodes, num_sampled_edges, L, i)
+            if trim is not None:
+                n_in, n_edges, n_out = trim
+                x = conv(x[:n_in], edge_index[:, :n_edges],
+                         num_target=n_out)
+            else:
+                x = conv(x, edge_index)
+            if i < L - 1:
                 x = F.relu(x)
                 x = F.dropout(x, p=self.dropout, training=self.training)
         return x
