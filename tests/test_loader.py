@@ -219,3 +219,26 @@ def test_feature_id2index_roundtrip():
     ids = torch.tensor([3, 7, 1])
     assert (f[ids] == ids.float().unsqueeze(1)).all()
     assert (f.cpu_get(ids) == ids.float().unsqueeze(1)).all()
+
+
+def test_hetero_edge_dir_in(ring_graph):
+    """'in' sampling over a hetero graph: results keyed by the reversed
+    edge type (reference neighbor_sampler.py:261-269 convention)."""
+    ei = ring_graph["edge_index"]
+    # graph stores (user, buys, item); with edge_dir='in' we walk item<-user
+    ds = Dataset(edge_dir="in")
+    ds.init_graph(edge_index={("user", "buys", "item"): ei},
+                  graph_mode="CPU", num_nodes=40)
+    ds.init_node_features({"user": ring_graph["feats"],
+                           "item": ring_graph["feats"]}, with_gpu=False)
+    loader = NeighborLoader(ds, [2], input_nodes=("item", torch.arange(10)),
+                            batch_size=5, edge_dir="in")
+    data = next(iter(loader))
+    rev = ("item", "rev_buys", "user")
+    assert rev in data.edge_types
+    ei_s = data[rev].edge_index
+    # item i's in-sources under 'buys' are users i-1, i-2
+    items = data["item"].node[ei_s[0]]
+    users = data["user"].node[ei_s[1]]
+    diff = (items - users) % 40
+    assert ((diff == 1) | (diff == 2)).all()
