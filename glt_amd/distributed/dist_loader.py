@@ -113,7 +113,12 @@ class DistLoader:
         if self._mode == "collocated":
             self._producer.reset()
         elif self._mode == "mp":
-            # drain stale messages, then start a new epoch
+            # drain stale messages from an abandoned epoch, then start anew
+            try:
+                while not self._channel.empty():
+                    self._channel.recv(timeout_ms=100)
+            except Exception:
+                pass
             self._producer.produce_all()
         else:
             from . import dist_client
@@ -142,6 +147,13 @@ class DistLoader:
                     raise StopIteration
                 continue
             return self._collate(msg)
+
+    def __len__(self):
+        n = len(self.input_data)
+        bs = self.sampling_config.batch_size
+        if self.sampling_config.drop_last:
+            return n // bs
+        return (n + bs - 1) // bs
 
     def _collate(self, msg):
         out, x, y, ea = decode_sample_message(msg)
