@@ -100,10 +100,9 @@ class _Prefetcher:
                 t.record_stream(cur)
             return t
 
-        from ..pygcompat.data import _apply
+        from ..pygcompat.data import apply_to_tensors
 
-        for k, v in list(data.items()) if hasattr(data, "items") else []:
-            data[k] = _apply(v, rec)
+        apply_to_tensors(data, rec)
         return data
 
     def stop(self):

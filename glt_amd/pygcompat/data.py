@@ -123,6 +123,20 @@ class Data:
         return f"Data({', '.join(parts)})"
 
 
+def apply_to_tensors(data, fn):
+    """Apply fn to every tensor held by a Data or HeteroData in place."""
+    if isinstance(data, HeteroData):
+        for s in list(data._node_stores.values()) + \
+                list(data._edge_stores.values()):
+            apply_to_tensors(s, fn)
+        for k, v in list(data._global.items()):
+            data._global[k] = _apply(v, fn)
+        return data
+    for k, v in list(data._store.items()):
+        data._store[k] = _apply(v, fn)
+    return data
+
+
 class _TypeStore(Data):
     """Per-node-type / per-edge-type store inside HeteroData."""
 
