@@ -238,10 +238,8 @@ class DistNeighborSampler:
     # ------------------------------------------------------------------
     async def _hetero_multihop(self, seed_dict, metadata=None
                                ) -> HeteroSamplerOutput:
-        from ..sampler.neighbor_sampler import _PyHeteroInducer
-
         s = self.sampler
-        inducer = _PyHeteroInducer(s._make_inducer)
+        inducer = s._acquire_hetero_inducer()
         frontier = inducer.init_node(seed_dict)
         out_nodes = {t: [v] for t, v in frontier.items()}
         num_nodes = {t: [v.numel()] for t, v in frontier.items()}
@@ -286,6 +284,7 @@ class DistNeighborSampler:
             n = inducer.nodes(t)
             batch[t] = n[: torch.unique(seed_dict[t]).numel()] \
                 if n is not None else seed_dict[t]
+        s._release_hetero_inducer(inducer)
         return HeteroSamplerOutput(
             node={t: torch.cat(v) for t, v in out_nodes.items()},
             row={k: torch.cat(v) for k, v in rows.items()},

@@ -41,7 +41,8 @@ class _PyHeteroInducer:
         return self._inducers[ntype]
 
     def init_node(self, seed_dict: Dict[NodeType, torch.Tensor]):
-        self._inducers = {}
+        # keep per-type inducers across batches (capacity retained);
+        # init_node on a native inducer resets its table
         self._uniq = {}
         out = {}
         for ntype, seeds in seed_dict.items():
@@ -97,6 +98,7 @@ class NeighborSampler(BaseSampler):
         self.is_hetero = isinstance(graph, dict)
         self._lock = threading.Lock()
         self._inducer_pool = []
+        self._hetero_pool = []
         if seed is not None:
             _C.manual_seed(seed)
         g0 = next(iter(graph.values())) if self.is_hetero else graph
@@ -127,6 +129,17 @@ class NeighborSampler(BaseSampler):
         with self._lock:
             if len(self._inducer_pool) < 8:
                 self._inducer_pool.append(ind)
+
+    def _acquire_hetero_inducer(self):
+        with self._lock:
+            if self._hetero_pool:
+                return self._hetero_pool.pop()
+        return _PyHeteroInducer(self._make_inducer)
+
+    def _release_hetero_inducer(self, ind):
+        with self._lock:
+            if len(self._hetero_pool) < 8:
+                self._hetero_pool.append(ind)
 
     def _seeds_to_device(self, seeds: torch.Tensor) -> torch.Tensor:
         return seeds.long().to(self._sample_device, non_blocking=True)
@@ -213,7 +226,7 @@ class NeighborSampler(BaseSampler):
 
     def _hetero_multihop(self, seed_dict: Dict[NodeType, torch.Tensor],
                          metadata=None) -> HeteroSamplerOutput:
-        inducer = _PyHeteroInducer(self._make_inducer)
+        inducer = self._acquire_hetero_inducer()
         frontier = inducer.init_node(seed_dict)
         out_nodes: Dict[NodeType, List[torch.Tensor]] = {
             t: [v] for t, v in frontier.items()}
@@ -270,6 +283,7 @@ class NeighborSampler(BaseSampler):
         edge = {et: torch.cat(v) for et, v in eids.items()} if eids else None
         batch = {t: self._uniq_of(seed_dict, inducer, t)
                  for t in seed_dict.keys()}
+        self._release_hetero_inducer(inducer)
         return HeteroSamplerOutput(
             node=node, row=row, col=col, edge=edge, batch=batch,
             num_sampled_nodes=num_nodes, num_sampled_edges=num_edges,
