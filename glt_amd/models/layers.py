@@ -141,8 +141,10 @@ class GATConv(nn.Module):
         # scatter softmax over tgt
         e_max = torch.full((nt, self.heads), float("-inf"),
                            device=e.device, dtype=e.dtype)
-        e_max.scatter_reduce_(0, tgt.unsqueeze(1).expand_as(e), e,
+        e_max.scatter_reduce_(0, tgt.unsqueeze(1).expand_as(e), e.detach(),
                               reduce="amax", include_self=True)
+        # softmax is shift-invariant: the max is a constant offset, so it
+        # carries no gradient (and its scatter-amax backward is skipped)
         e = (e - e_max.index_select(0, tgt)).exp()
         denom = torch.zeros(nt, self.heads, device=e.device, dtype=e.dtype)
         denom.index_add_(0, tgt, e)
