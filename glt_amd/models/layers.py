@@ -65,10 +65,13 @@ class SAGEConv(nn.Module):
             agg = x.new_zeros(n, x.size(1))
             agg.index_add_(0, tgt, x.index_select(0, src))
             agg = agg / _degree(tgt, n).unsqueeze(1).to(x.dtype)
-        out = self.lin_l(agg)
         if self.lin_r is not None:
-            out = out + self.lin_r(x[:n])
-        return out
+            # one [n, 2F] x [2F, out] GEMM instead of two skinny K=F GEMMs
+            # (halves GEMM launches; wgrad reduces once over n)
+            w = torch.cat([self.lin_l.weight, self.lin_r.weight], dim=1)
+            return F.linear(torch.cat([agg, x[:n]], dim=1), w,
+                            self.lin_l.bias)
+        return self.lin_l(agg)
 
 
 class GCNConv(nn.Module):
