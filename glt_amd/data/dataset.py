@@ -104,6 +104,24 @@ class Dataset:
             self.node_labels = node_label_data
         return self
 
+    def load_vineyard(self, vineyard_id: str, vineyard_socket: str,
+                      edges, edge_weights=None, node_features=None,
+                      edge_features=None, node_labels=None):
+        """Load a GraphScope (v6d) fragment (capability parity: reference
+        data/dataset.py:155-234).  Requires the vineyard runtime; see
+        glt_amd.data.vineyard_utils."""
+        from . import vineyard_utils as v6d
+
+        for etype in edges:
+            indptr, indices, eids = v6d.vineyard_to_csr(
+                vineyard_socket, vineyard_id, etype[0] if
+                isinstance(etype, tuple) else etype,
+                etype[1] if isinstance(etype, tuple) else etype,
+                self.edge_dir)
+            self.init_graph(edge_index=(indptr, indices), edge_ids=eids,
+                            layout="CSR")
+        return self
+
     def random_node_split(self, num_val: Union[int, float],
                           num_test: Union[int, float],
                           ntype: Optional[NodeType] = None):

@@ -41,12 +41,26 @@ class DistDataset(Dataset):
              graph_mode: str = "ZERO_COPY",
              input_layout: str = "COO",
              feature_with_gpu: bool = True,
+             graph_caching: bool = False,
              device_group_list: Optional[List[DeviceGroup]] = None,
              whole_node_label_file: Optional[Union[str, Dict]] = None,
              device: Optional[int] = None):
-        """Load one partition saved in the GLT on-disk layout."""
+        """Load one partition saved in the GLT on-disk layout.
+
+        graph_caching=True loads the whole-topology cache written by
+        `partition.save_graph_cache` (every rank holds the full graph;
+        only features stay partitioned — reference graph_caching mode).
+        """
         (num_parts, graph_data, node_feat_data, edge_feat_data, node_pb,
          edge_pb) = load_partition(root_dir, partition_idx)
+        if graph_caching:
+            import os as _os
+
+            from ..partition.base import _load_graph_dir
+
+            whole = _load_graph_dir(_os.path.join(root_dir, "graph"))
+            if whole is not None:
+                graph_data = whole
         self.num_partitions = num_parts
         self.partition_idx = partition_idx
         self.node_pb = node_pb
