@@ -297,3 +297,68 @@ def test_sample_prob_gpu(ring_graph):
     prob = s.sample_prob(torch.tensor([0], device="cuda"), 40)
     assert abs(prob[1].item() - 0.5) < 1e-4
     assert prob[0].item() == 1.0
+
+
+@pytest.mark.timeout(300)
+def test_end_to_end_learning_sbm():
+    """Training correctness through the whole GPU pipeline: a stochastic
+    block-model graph whose labels are the communities; 3-hop GraphSAGE on
+    noisy community features must reach high train accuracy."""
+    from glt_amd import Dataset, NeighborLoader
+    from glt_amd.models import GraphSAGE
+
+    glt_amd.seed_everything(0)
+    n, k = 20_000, 10
+    comm = torch.randint(0, k, (n,))
+    # intra-community edges (90%) + noise edges (10%)
+    e = n * 20
+    src = torch.randint(0, n, (e,))
+    same = torch.rand(e) < 0.9
+    dst = torch.where(
+        same,
+        # random node of the same community: offset walk within community
+        (src + torch.randint(1, n, (e,))) % n,
+        torch.randint(0, n, (e,)))
+    # force same-community dst for the "same" edges by re-mapping through a
+    # community-sorted permutation
+    order = torch.argsort(comm)
+    rank_of = torch.empty(n, dtype=torch.long)
+    rank_of[order] = torch.arange(n)
+    counts = torch.bincount(comm, minlength=k)
+    starts = torch.zeros(k, dtype=torch.long)
+    torch.cumsum(counts, 0, out=starts[1:] if k > 1 else starts)
+    starts = torch.cat([torch.zeros(1, dtype=torch.long),
+                        torch.cumsum(counts, 0)[:-1]])
+    rnd = torch.rand(e)
+    same_dst = order[(starts[comm[src]] +
+                      (rnd * counts[comm[src]].float()).long().clamp(
+                          max=counts.max() - 1).clamp(min=0)) % n]
+    dst = torch.where(same, same_dst, dst)
+    feats = torch.nn.functional.one_hot(comm, k).float()
+    feats = feats + 0.5 * torch.randn(n, k)
+    ds = Dataset()
+    ds.init_graph(edge_index=torch.stack([src, dst]), graph_mode="CUDA",
+                  num_nodes=n, device=0)
+    ds.init_node_features(feats, split_ratio=1.0, device=0)
+    ds.init_node_labels(comm.cuda())
+    dev = torch.device("cuda", 0)
+    loader = NeighborLoader(ds, [10, 5], input_nodes=torch.arange(n),
+                            batch_size=1024, shuffle=True, device=dev,
+                            to_device=dev, prefetch=2)
+    model = GraphSAGE(k, 64, 2, out_channels=k).to(dev)
+    opt = torch.optim.Adam(model.parameters(), lr=5e-3)
+    correct = total = 0
+    for epoch in range(3):
+        for data in loader:
+            opt.zero_grad(set_to_none=True)
+            out = model(data.x, data.edge_index, data.num_sampled_nodes,
+                        data.num_sampled_edges)[:data.batch_size]
+            y = data.y[:data.batch_size]
+            loss = torch.nn.functional.cross_entropy(out, y)
+            loss.backward()
+            opt.step()
+            if epoch == 2:
+                correct += int((out.argmax(-1) == y).sum())
+                total += y.numel()
+    acc = correct / max(total, 1)
+    assert acc > 0.85, f"end-to-end accuracy too low: {acc}"
