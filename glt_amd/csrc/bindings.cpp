@@ -188,6 +188,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("rows", &FeatureStorePy::rows)
       .def("dim", &FeatureStorePy::dim);
 
+  m.def("sage_gemm", &hip_sage_gemm, py::arg("A"), py::arg("B"),
+        py::arg("bias") = py::none());
   m.def("segment_mean_fwd", &hip_segment_mean_fwd);
   m.def("segment_mean_bwd", &hip_segment_mean_bwd);
 

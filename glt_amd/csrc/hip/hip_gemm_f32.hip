@@ -1,0 +1,162 @@
+/* Hand-written f32 MFMA GEMM for GNN projection shapes (gfx950).
+ *
+ * C[m,n] = A[m,k] @ B[k,n] (+ bias), fp32 in / fp32 out, exact f32
+ * numerics via v_mfma_f32_16x16x4_f32 (the CDNA4 "SGEMM" matrix-core op:
+ * 157 TF chip peak = 16x the rocBLAS pick we measured on these shapes —
+ * tall-skinny m~1e5, n=256, k=200/512 batches where Tensile chooses
+ * MT32x32 tiles at ~51 TF).
+ *
+ * Geometry: 64x64 block tile, 4 waves as 2x2, each wave a 32x32 tile of
+ * four 16x16 fragments (4 independent accumulators: the 16x16x4 form needs
+ * >=2 for back-to-back issue, guide §3).  K staged through LDS in BK=16
+ * steps with +1-row padding (bank-conflict rule, guide §6 G4); edge tiles
+ * (m tail, k tail) zero-filled so no host-side padding is needed.
+ */
+#include "hip_common.h"
+#include "../include/common.h"
+
+namespace glt {
+
+namespace {
+
+using f32x4 = __attribute__((__vector_size__(4 * sizeof(float)))) float;
+
+constexpr int BM = 64, BN = 64, BK = 16;
+constexpr int PAD_A = 1;  // As[BM][BK+1]
+constexpr int PAD_B = 1;  // Bs[BK][BN+1]
+
+__global__ __launch_bounds__(256)
+void sage_gemm_f32_kernel(const float* __restrict__ A,
+                          const float* __restrict__ B,
+                          const float* __restrict__ bias,
+                          float* __restrict__ C,
+                          int64_t M, int64_t K, int64_t N) {
+  __shared__ float As[BM][BK + PAD_A];
+  __shared__ float Bs[BK][BN + PAD_B];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;          // 0..3
+  const int wr = wave >> 1;           // wave row 0..1 (32 rows each)
+  const int wc = wave & 1;            // wave col 0..1 (32 cols each)
+
+  const int64_t block_row = (int64_t)blockIdx.x * BM;
+  const int64_t block_col = (int64_t)blockIdx.y * BN;
+
+  f32x4 acc[2][2] = {};  // 2x2 fragments of 16x16
+
+  // cooperative staging indices: 256 threads, A tile 64x16 = 1024 floats
+  // -> 4 per thread (one row of 4); B tile 16x64 -> 4 per thread.
+  const int a_row = tid >> 2;                 // 0..63
+  const int a_col4 = (tid & 3) * 4;           // 0,4,8,12
+  const int b_row = tid >> 4;                 // 0..15
+  const int b_col4 = (tid & 15) * 4;          // 0..60
+
+  const int frag_i = lane & 15;               // row within 16x16 A frag
+  const int frag_k = lane >> 4;               // k lane 0..3
+
+  for (int64_t k0 = 0; k0 < K; k0 += BK) {
+    // --- stage A[block_row..+64, k0..+16] ---
+    {
+      const int64_t g_row = block_row + a_row;
+      float v0 = 0.f, v1 = 0.f, v2 = 0.f, v3 = 0.f;
+      if (g_row < M) {
+        const int64_t base = g_row * K + k0 + a_col4;
+        const int64_t kmax = K - k0;
+        if (a_col4 + 3 < kmax) {
+          const float4 f = *reinterpret_cast<const float4*>(&A[base]);
+          v0 = f.x; v1 = f.y; v2 = f.z; v3 = f.w;
+        } else {
+          if (a_col4 + 0 < kmax) v0 = A[base + 0];
+          if (a_col4 + 1 < kmax) v1 = A[base + 1];
+          if (a_col4 + 2 < kmax) v2 = A[base + 2];
+          if (a_col4 + 3 < kmax) v3 = A[base + 3];
+        }
+      }
+      As[a_row][a_col4 + 0] = v0;
+      As[a_row][a_col4 + 1] = v1;
+      As[a_row][a_col4 + 2] = v2;
+      As[a_row][a_col4 + 3] = v3;
+    }
+    // --- stage B[k0..+16, block_col..+64] ---
+    {
+      const int64_t g_k = k0 + b_row;
+      float v0 = 0.f, v1 = 0.f, v2 = 0.f, v3 = 0.f;
+      if (g_k < K) {
+        const int64_t base = g_k * N + block_col + b_col4;
+        const float4 f = *reinterpret_cast<const float4*>(&B[base]);
+        v0 = f.x; v1 = f.y; v2 = f.z; v3 = f.w;
+      }
+      Bs[b_row][b_col4 + 0] = v0;
+      Bs[b_row][b_col4 + 1] = v1;
+      Bs[b_row][b_col4 + 2] = v2;
+      Bs[b_row][b_col4 + 3] = v3;
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int kk = 0; kk < BK; kk += 4) {
+      // wave's A rows: wr*32 + fi*16 + frag_i ; B cols: wc*32 + fj*16 + ...
+      float a0 = As[wr * 32 + frag_i][kk + frag_k];
+      float a1 = As[wr * 32 + 16 + frag_i][kk + frag_k];
+      float b0 = Bs[kk + frag_k][wc * 32 + frag_i];
+      float b1 = Bs[kk + frag_k][wc * 32 + 16 + frag_i];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc[0][0],
+                                                       0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc[0][1],
+                                                       0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc[1][0],
+                                                       0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc[1][1],
+                                                       0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // Epilogue: C/D fragment layout for 16x16x4f32 is col = lane&15,
+  // row = (lane>>4)*4 + reg (guide §3).
+  const int c_col = lane & 15;
+  const int c_row0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int fi = 0; fi < 2; ++fi) {
+#pragma unroll
+    for (int fj = 0; fj < 2; ++fj) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int64_t row = block_row + wr * 32 + fi * 16 + c_row0 + r;
+        const int64_t col = block_col + wc * 32 + fj * 16 + c_col;
+        if (row < M) {
+          float v = acc[fi][fj][r];
+          if (bias != nullptr) v += bias[col];
+          C[row * N + col] = v;
+        }
+      }
+    }
+  }
+}
+
+}  // namespace
+
+// C = A @ B (+bias). A [M,K] row-major, B [K,N] row-major; N % 64 == 0.
+torch::Tensor hip_sage_gemm(const torch::Tensor& A, const torch::Tensor& B,
+                            const c10::optional<torch::Tensor>& bias) {
+  TORCH_CHECK(A.is_cuda() && B.is_cuda(), "device tensors required");
+  TORCH_CHECK(A.scalar_type() == torch::kFloat32 &&
+              B.scalar_type() == torch::kFloat32, "fp32 only");
+  TORCH_CHECK(A.dim() == 2 && B.dim() == 2 && A.size(1) == B.size(0));
+  const int64_t M = A.size(0), K = A.size(1), N = B.size(1);
+  TORCH_CHECK(N % 64 == 0, "N must be a multiple of 64");
+  auto Ac = A.contiguous();
+  auto Bc = B.contiguous();
+  auto C = torch::empty({M, N}, A.options());
+  if (M == 0) return C;
+  dim3 grid((uint32_t)((M + BM - 1) / BM), (uint32_t)(N / BN));
+  hipLaunchKernelGGL(sage_gemm_f32_kernel, grid, dim3(256), 0,
+                     current_stream(), Ac.data_ptr<float>(),
+                     Bc.data_ptr<float>(),
+                     bias.has_value() ? bias->data_ptr<float>() : nullptr,
+                     C.data_ptr<float>(), M, K, N);
+  return C;
+}
+
+}  // namespace glt

@@ -70,6 +70,10 @@ torch::Tensor hip_segment_mean_bwd(const torch::Tensor& dy,
                                    const torch::Tensor& offsets,
                                    int64_t n_src);
 
+// --- f32 MFMA projection GEMM (hip_gemm_f32.hip) ----------------------------
+torch::Tensor hip_sage_gemm(const torch::Tensor& A, const torch::Tensor& B,
+                            const c10::optional<torch::Tensor>& bias);
+
 // --- memory plumbing (hip_mem.hip) -----------------------------------------
 // Device-dtype alias of (pinned/registered) host memory; keeps `src` alive.
 torch::Tensor host_mapped_view(const torch::Tensor& src, int64_t device_index);

@@ -362,3 +362,27 @@ def test_end_to_end_learning_sbm():
                 total += y.numel()
     acc = correct / max(total, 1)
     assert acc > 0.85, f"end-to-end accuracy too low: {acc}"
+
+
+def test_mfma_gemm_numerics():
+    from glt_amd.ops import mfma_linear
+
+    torch.manual_seed(0)
+    for m, k, n in ((1000, 200, 256), (64, 16, 64), (130, 100, 128),
+                    (4096, 512, 256), (77, 7, 64)):
+        x = torch.randn(m, k, device="cuda", requires_grad=True)
+        w = torch.randn(n, k, device="cuda", requires_grad=True)
+        b = torch.randn(n, device="cuda", requires_grad=True)
+        out = mfma_linear(x, w, b)
+        ref = torch.nn.functional.linear(x.detach(), w.detach(), b.detach())
+        assert torch.allclose(out, ref, atol=1e-3, rtol=1e-4), \
+            (m, k, n, (out - ref).abs().max().item())
+        g = torch.randn_like(out)
+        out.backward(g)
+        xr = x.detach().clone().requires_grad_()
+        wr = w.detach().clone().requires_grad_()
+        br = b.detach().clone().requires_grad_()
+        torch.nn.functional.linear(xr, wr, br).backward(g)
+        assert torch.allclose(x.grad, xr.grad, atol=1e-3, rtol=1e-4)
+        assert torch.allclose(w.grad, wr.grad, atol=1e-2, rtol=1e-4)
+        assert torch.allclose(b.grad, br.grad, atol=1e-2, rtol=1e-4)
