@@ -68,9 +68,13 @@ class SAGEConv(nn.Module):
         if self.lin_r is not None:
             # one [n, 2F] x [2F, out] GEMM instead of two skinny K=F GEMMs
             # (halves GEMM launches; wgrad reduces once over n)
+            from ..ops import mfma_linear, use_mfma_linear
+
             w = torch.cat([self.lin_l.weight, self.lin_r.weight], dim=1)
-            return F.linear(torch.cat([agg, x[:n]], dim=1), w,
-                            self.lin_l.bias)
+            xin = torch.cat([agg, x[:n]], dim=1)
+            if use_mfma_linear(xin, w):
+                return mfma_linear(xin, w, self.lin_l.bias)
+            return F.linear(xin, w, self.lin_l.bias)
         return self.lin_l(agg)
 
 

@@ -44,6 +44,10 @@ def mfma_linear(x: torch.Tensor, weight: torch.Tensor,
 
 
 def use_mfma_linear(x: torch.Tensor, weight: torch.Tensor) -> bool:
+    """Dispatch policy from MI355X measurements: the custom
+    kernel ~doubles rocBLAS on narrow outputs (n<=128: Tensile picks a
+    32-wide tile at ~32 TF there) and ties it at n=256, where we keep the
+    library."""
     return (x.is_cuda and x.dtype == torch.float32
             and weight.dtype == torch.float32
-            and weight.size(0) % 64 == 0)
+            and weight.size(0) % 64 == 0 and weight.size(0) <= 128)
