@@ -85,3 +85,124 @@ def test_dist_sampling_two_ranks_one_gpu():
             p.terminate()
     for rank, err in results:
         assert err is None, f"rank {rank}:\n{err}"
+
+
+def _worker_mp_gpu(rank, world, port, fail_q):
+    try:
+        import torch
+
+        import glt_amd
+        from glt_amd.data import Feature
+        from glt_amd.distributed import (DistDataset, DistNeighborLoader,
+                                         MpDistSamplingWorkerOptions,
+                                         init_worker_group)
+        from glt_amd.partition import GLTPartitionBook
+
+        glt_amd.seed_everything(5 + rank)
+        torch.cuda.set_device(0)
+        init_worker_group(world, rank)
+        rows, cols = [], []
+        for v in range(VNUM):
+            rows += [v, v]
+            cols += [(v + 1) % VNUM, (v + 2) % VNUM]
+        ds = DistDataset(num_partitions=1, partition_idx=0)
+        ds.init_graph(edge_index=torch.tensor([rows, cols]),
+                      graph_mode="CPU", num_nodes=VNUM)
+        ds.node_pb = GLTPartitionBook(torch.zeros(VNUM, dtype=torch.uint8))
+        feats = torch.arange(VNUM, dtype=torch.float32).unsqueeze(1) \
+            .repeat(1, 16)
+        # GPU-tiered feature store crosses the process boundary via its
+        # shared host tensor and lazily re-materializes on the worker GPU
+        ds.node_features = Feature(feats, split_ratio=1.0, device=0,
+                                   with_gpu=True)
+        ds.node_labels = torch.arange(VNUM)
+        opts = MpDistSamplingWorkerOptions(
+            num_workers=2, master_addr="127.0.0.1", master_port=port,
+            channel_size="32MB", channel_capacity=16, pin_memory=True)
+        loader = DistNeighborLoader(ds, [2, 2],
+                                    input_nodes=torch.arange(VNUM),
+                                    batch_size=5,
+                                    to_device=torch.device("cuda", 0),
+                                    worker_options=opts)
+        n = 0
+        for data in loader:
+            node = data.node.cpu()
+            ei = data.edge_index.cpu()
+            diff = (node[ei[1]] - node[ei[0]]) % VNUM
+            assert ((diff == 1) | (diff == 2)).all()
+            assert (data.x.cpu() == node.float().unsqueeze(1)).all()
+            n += 1
+        assert n == 8, n
+        loader.shutdown()
+        fail_q.put((rank, None))
+    except Exception:  # noqa: BLE001
+        import traceback
+
+        fail_q.put((rank, traceback.format_exc()))
+
+
+@pytest.mark.timeout(300)
+def test_mp_sampling_workers_gpu_features():
+    from glt_amd.utils import get_free_port
+
+    ctx = mp.get_context("spawn")
+    port = get_free_port()
+    q = ctx.Queue()
+    p = ctx.Process(target=_worker_mp_gpu, args=(0, 1, port, q))
+    p.start()
+    rank, err = q.get(timeout=280)
+    p.join(timeout=30)
+    if p.is_alive():
+        p.terminate()
+    assert err is None, err
+
+
+@pytest.mark.timeout(300)
+def test_memory_stable_over_steps():
+    """Leak check: steady-state sampling+training must not grow VRAM."""
+    import torch
+
+    import glt_amd
+    from glt_amd import Dataset, NeighborLoader
+    from glt_amd.models import GraphSAGE
+
+    glt_amd.seed_everything(0)
+    n = 100_000
+    src = torch.randint(0, n, (2_000_000,))
+    dst = torch.randint(0, n, (2_000_000,))
+    ds = Dataset()
+    ds.init_graph(edge_index=torch.stack([src, dst]), graph_mode="CUDA",
+                  num_nodes=n, device=0)
+    ds.init_node_features(torch.randn(n, 64), split_ratio=1.0, device=0)
+    ds.init_node_labels(torch.randint(0, 10, (n,)).cuda())
+    dev = torch.device("cuda", 0)
+    loader = NeighborLoader(ds, [10, 5], input_nodes=torch.arange(n),
+                            batch_size=1024, shuffle=True, device=dev,
+                            to_device=dev, prefetch=2)
+    model = GraphSAGE(64, 128, 2, out_channels=10).to(dev)
+    opt = torch.optim.Adam(model.parameters())
+    it = iter(loader)
+
+    def step():
+        nonlocal it
+        try:
+            data = next(it)
+        except StopIteration:
+            it = iter(loader)
+            data = next(it)
+        opt.zero_grad(set_to_none=True)
+        out = model(data.x, data.edge_index, data.num_sampled_nodes,
+                    data.num_sampled_edges)[:data.batch_size]
+        torch.nn.functional.cross_entropy(
+            out, data.y[:data.batch_size]).backward()
+        opt.step()
+
+    for _ in range(30):
+        step()
+    torch.cuda.synchronize()
+    base = torch.cuda.memory_allocated()
+    for _ in range(150):
+        step()
+    torch.cuda.synchronize()
+    grow = torch.cuda.memory_allocated() - base
+    assert grow < 256 * (1 << 20), f"VRAM grew {grow / (1 << 20):.1f} MiB"
