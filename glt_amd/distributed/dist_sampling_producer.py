@@ -80,11 +80,24 @@ def _sampling_worker_loop(worker_idx: int, dataset: DistDataset,
                 else:
                     raise ValueError(sampling_config.sampling_type)
                 futures.append(fut)
+            errors = []
             for f in futures:
-                f.result()
+                try:
+                    f.result()
+                except Exception:  # noqa: BLE001
+                    import traceback
+
+                    errors.append(traceback.format_exc())
+            # END must flow even on failure or the trainer hangs forever
             channel.send({END_KEY: torch.tensor([1])})
             with done_counter.get_lock():
                 done_counter.value += 1
+            if errors:
+                import sys
+
+                print(f"[glt_amd sampling worker {worker_idx}] "
+                      f"{len(errors)} batch(es) failed:\n{errors[0]}",
+                      file=sys.stderr, flush=True)
         sampler.shutdown()
         shutdown_rpc()
     except KeyboardInterrupt:

@@ -148,6 +148,8 @@ class NeighborSampler(BaseSampler):
                        etype: Optional[EdgeType] = None) -> NeighborOutput:
         """Uniform (or weighted) one-hop sample from srcs."""
         g = self.graph[etype] if etype is not None else self.graph
+        # sampling runs where the CSR lives, whatever device srcs arrive on
+        srcs = srcs.to(g.indptr.device, non_blocking=True)
         weighted = self.with_weight and g.edge_weights is not None
         nbrs, num, eids = self._C.sample_neighbors(
             g.indptr, g.indices, srcs, k,
