@@ -40,6 +40,11 @@ class DistLoader:
         self._producer = None
         self._ends_seen = 0
         self._epoch = 0
+        # zero-copy safety: decoded messages alias the shm ring; their
+        # blocks must stay referenced until the async H2D copies complete
+        import collections as _c
+
+        self._inflight = _c.deque()
 
         if isinstance(self.worker_options,
                       CollocatedDistSamplingWorkerOptions):
@@ -164,7 +169,14 @@ class DistLoader:
                                   node_feat_dict=x, edge_feat_dict=ea,
                                   edge_dir=self.edge_dir)
         if self.to_device is not None:
+            on_gpu = (torch.device(self.to_device).type == "cuda")
             data = data.to(self.to_device, non_blocking=True)
+            if on_gpu:
+                ev = torch.cuda.Event()
+                ev.record()
+                self._inflight.append((msg, ev))
+                while self._inflight and self._inflight[0][1].query():
+                    self._inflight.popleft()
         return data
 
     def shutdown(self):
