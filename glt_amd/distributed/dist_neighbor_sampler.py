@@ -137,7 +137,9 @@ class DistNeighborSampler:
     async def _sample_one_hop(self, srcs: torch.Tensor, k: int,
                               etype: Optional[EdgeType] = None
                               ) -> NeighborOutput:
-        device = self.sampler.device
+        # the whole sampling/induction path lives on the graph's device
+        # (CPU for CPU-mode graphs even when features are GPU-tiered)
+        device = self.sampler._sample_device
         srcs = srcs.to(device)
         if self.data.num_partitions <= 1:
             return self.sampler.sample_one_hop(srcs, k, etype=etype)
@@ -196,7 +198,7 @@ class DistNeighborSampler:
     async def _multihop(self, seeds: torch.Tensor, metadata=None
                         ) -> SamplerOutput:
         inducer = self.sampler._acquire_inducer()
-        uniq = inducer.init_node(seeds.to(self.sampler.device))
+        uniq = inducer.init_node(seeds.to(self.sampler._sample_device))
         out_nodes, num_nodes, num_edges = [uniq], [uniq.numel()], []
         rows, cols, eids = [], [], []
         srcs = uniq
@@ -211,7 +213,7 @@ class DistNeighborSampler:
             if out.edge is not None:
                 eids.append(out.edge)
             srcs = nodes
-        dev = self.sampler.device
+        dev = self.sampler._sample_device
         self.sampler._release_inducer(inducer)
         return SamplerOutput(
             node=torch.cat(out_nodes),
@@ -312,8 +314,8 @@ class DistNeighborSampler:
             # delegate to local hetero link sampling then collect
             out = s.sample_from_edges(inputs)
             return await self._collect(out)
-        row = inputs.row.to(s.device)
-        col = inputs.col.to(s.device)
+        row = inputs.row.to(s._sample_device)
+        col = inputs.col.to(s._sample_device)
         num_pos = row.numel()
         g = s.graph
         from .. import _C
@@ -333,9 +335,9 @@ class DistNeighborSampler:
                 torch.cat([local[num_pos:2 * num_pos],
                            local[2 * num_pos + n_neg:]])])
             label = torch.cat([
-                inputs.label.to(s.device) if inputs.label is not None
-                else torch.ones(num_pos, device=s.device),
-                torch.zeros(n_neg, device=s.device)])
+                inputs.label.to(row.device) if inputs.label is not None
+                else torch.ones(num_pos, device=row.device),
+                torch.zeros(n_neg, device=row.device)])
             out.metadata = {"edge_label_index": eli, "edge_label": label}
         elif neg is not None and neg.is_triplet():
             amount = int(neg.amount)
@@ -357,7 +359,7 @@ class DistNeighborSampler:
             out.metadata = {
                 "edge_label_index": torch.stack([local[:num_pos],
                                                  local[num_pos:]]),
-                "edge_label": inputs.label.to(s.device)
+                "edge_label": inputs.label.to(row.device)
                 if inputs.label is not None else None}
         return await self._collect(out)
 
