@@ -156,7 +156,10 @@ class Graph:
             def mapped(t):
                 return None if t is None else _C.host_mapped_view(t, dev)
 
-            self._indptr = mapped(topo.indptr)
+            # indptr is (N+1)*8 bytes — tiny next to indices — and is read
+            # twice per seed (degree + base): keep it in HBM so only the
+            # neighbor-list gathers ride UVA/PCIe
+            self._indptr = topo.indptr.to(torch.device("cuda", dev))
             self._indices = mapped(topo.indices)
             self._edge_ids = mapped(topo.edge_ids)
             self._edge_weights = mapped(topo.edge_weights)
