@@ -18,6 +18,7 @@ from .dist_server import (DistServer, get_server, init_server,
 from .dist_client import (async_request_server, init_client,
                           request_server, shutdown_client)
 from .event_loop import ConcurrentEventLoop
+from .pyg_remote_backend import RemoteFeatureStore, RemoteGraphStore
 from .message import decode_sample_message, encode_sampler_output
 from .rpc import (barrier, init_rpc, rpc_is_initialized, rpc_register,
                   rpc_request_async, shutdown_rpc)
@@ -32,6 +33,7 @@ __all__ = [
     "DistMpSamplingProducer", "DistServer", "get_server", "init_server",
     "wait_and_shutdown_server", "async_request_server", "init_client",
     "request_server", "shutdown_client", "ConcurrentEventLoop",
+    "RemoteFeatureStore", "RemoteGraphStore",
     "decode_sample_message", "encode_sampler_output", "barrier",
     "init_rpc", "rpc_is_initialized", "rpc_register", "rpc_request_async",
     "shutdown_rpc",

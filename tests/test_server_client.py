@@ -72,6 +72,16 @@ def _client_proc(port, q):
                              torch.tensor([1, 2]))
         assert pid.tolist() == [0, 0]
 
+        from glt_amd.distributed import RemoteFeatureStore, RemoteGraphStore
+
+        fs = RemoteFeatureStore()
+        assert (fs.get_tensor(torch.tensor([4])) ==
+                torch.full((1, 8), 4.0)).all()
+        assert fs.get_tensor(torch.tensor([9]), attr="y").tolist() == [9]
+        gs = RemoteGraphStore()
+        ei = gs.get_edge_index()
+        assert ei.size(1) == 2 * VNUM
+
         # remote sampling loader
         opts = RemoteDistSamplingWorkerOptions(
             server_rank=0, num_workers=2, buffer_size="8MB",
