@@ -47,6 +47,10 @@ def parse_args():
     p.add_argument("--amp", action="store_true",
                    help="bf16 autocast for model math (default fp32, "
                         "matching the reference)")
+    p.add_argument("--feature-mode", type=str, default="replicated",
+                   choices=["replicated", "xgmi-shard"],
+                   help="xgmi-shard: features sharded across ranks' HBM, "
+                        "gathered over hip-IPC peer pointers (xGMI)")
     return p.parse_args()
 
 
@@ -132,11 +136,15 @@ def main():
 
     ds = Dataset()
     ds.graph = graph
-    if has_gpu:
+    if has_gpu and args.feature_mode == "xgmi-shard" and world > 1:
+        from glt_amd.data import XgmiShardedFeature
+
+        ds.node_features = XgmiShardedFeature(
+            feats.cpu() if feats.is_cuda else feats, device=device.index)
+    elif has_gpu:
         # features fully HBM-resident (288 GB): split_ratio 1.0
         f = Feature(feats.cpu() if feats.is_cuda else feats,
                     split_ratio=1.0, device=device.index, with_gpu=True)
-        # keep the on-device copy we already have instead of re-uploading
         ds.node_features = f
     else:
         ds.node_features = Feature(feats, with_gpu=False)
@@ -226,6 +234,7 @@ def main():
                 "classes": args.classes,
                 "graph_mode": args.graph_mode,
                 "parallelism": f"dp{world}",
+                "feature_mode": args.feature_mode,
                 "epoch_time_s_equiv": round(
                     epoch_batches / (batches_per_sec / world), 3),
             },
