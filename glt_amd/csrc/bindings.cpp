@@ -190,6 +190,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
 
   m.def("sage_gemm", &hip_sage_gemm, py::arg("A"), py::arg("B"),
         py::arg("bias") = py::none());
+  m.def("gat_fused_fwd", &hip_gat_fused_fwd);
+  m.def("gat_fused_bwd", &hip_gat_fused_bwd);
   m.def("segment_mean_fwd", &hip_segment_mean_fwd);
   m.def("segment_mean_bwd", &hip_segment_mean_bwd);
 

@@ -140,6 +140,21 @@ class GATConv(nn.Module):
         h = h_src
         alpha_src = (h_src * self.att_src).sum(-1)
         alpha_dst = (h_tgt * self.att_dst).sum(-1)
+        if (getattr(self, "use_fused", True) and h_src.is_cuda
+                and h_src.dtype == torch.float32
+                and self.out_channels <= 128
+                and not (self.training and self.dropout > 0)):
+            # fused segment-softmax-aggregate (edges sorted by target)
+            from ..ops import gat_softmax_aggregate
+
+            out = gat_softmax_aggregate(h_src, alpha_src,
+                                        alpha_dst[:nt], tgt, src, nt,
+                                        self.negative_slope)
+            out = out.reshape(nt, self.heads * self.out_channels) \
+                if self.concat else out.mean(dim=1)
+            if self.bias is not None:
+                out = out + self.bias
+            return out
         # index_select (not advanced indexing): its backward is an
         # index_add scatter, avoiding the radix-sort the indexing backward
         # performs per gather (24 device sorts/step in RGAT otherwise)
