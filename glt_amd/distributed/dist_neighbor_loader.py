@@ -6,7 +6,7 @@ from typing import List, Optional, Union
 import torch
 
 from ..sampler import (EdgeSamplerInput, NegativeSampling, NodeSamplerInput,
-                       SamplingConfig, SamplingType)
+                       RemoteSamplerInput, SamplingConfig, SamplingType)
 from .dist_dataset import DistDataset
 from .dist_loader import DistLoader
 
@@ -20,7 +20,9 @@ class DistNeighborLoader(DistLoader):
                  collect_features: bool = True,
                  to_device: Optional[torch.device] = None,
                  worker_options=None):
-        if isinstance(input_nodes, tuple) and not isinstance(
+        if isinstance(input_nodes, RemoteSamplerInput):
+            inp = input_nodes  # server resolves the seed split/path
+        elif isinstance(input_nodes, tuple) and not isinstance(
                 input_nodes, NodeSamplerInput):
             inp = NodeSamplerInput(node=torch.as_tensor(input_nodes[1]),
                                    input_type=input_nodes[0])

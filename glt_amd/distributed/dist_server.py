@@ -90,6 +90,10 @@ class DistServer:
                                  num_workers: int, buffer_capacity: int,
                                  buffer_bytes: int, worker_key: str,
                                  worker_concurrency: int = 4) -> int:
+        from ..sampler import RemoteSamplerInput
+
+        if isinstance(seeds_input, RemoteSamplerInput):
+            seeds_input = seeds_input.resolve(self.dataset)
         with self._lock:
             pid = self._next_id
             self._next_id += 1

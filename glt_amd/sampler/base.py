@@ -189,6 +189,36 @@ class SamplingConfig:
     seed_stride: int = 1  # random-walk length when RANDOM_WALK
 
 
+@dataclass
+class RemoteSamplerInput:
+    """Seed specification resolved on the SERVER side (server-client mode):
+    either a named dataset split ('train'/'val'/'test') or a .pt file path
+    readable by the server (parity: reference sampler/base.py
+    RemoteSamplerInput semantics).
+    """
+    split: Optional[str] = None
+    path: Optional[str] = None
+    input_type: Optional[NodeType] = None
+
+    def resolve(self, dataset) -> "NodeSamplerInput":
+        import torch as _torch
+
+        if self.path:
+            seeds = _torch.load(self.path, weights_only=False)
+        else:
+            attr = {"train": "train_idx", "val": "val_idx",
+                    "test": "test_idx"}[self.split or "train"]
+            seeds = getattr(dataset, attr)
+            if seeds is None:
+                raise ValueError(
+                    f"dataset has no '{self.split}' split on the server")
+        return NodeSamplerInput(node=_torch.as_tensor(seeds),
+                                input_type=self.input_type)
+
+    def __len__(self):  # unknown client-side; producers resolve remotely
+        return 0
+
+
 class BaseSampler:
     """Abstract sampler interface (parity: reference sampler/base.py:444+)."""
 

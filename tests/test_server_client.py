@@ -28,6 +28,7 @@ def _make_dataset():
     from glt_amd.partition import GLTPartitionBook
 
     ds.node_pb = GLTPartitionBook(torch.zeros(VNUM, dtype=torch.uint8))
+    ds.train_idx = torch.arange(VNUM)  # server-resolved seed split
     return ds
 
 
@@ -82,12 +83,15 @@ def _client_proc(port, q):
         ei = gs.get_edge_index()
         assert ei.size(1) == 2 * VNUM
 
-        # remote sampling loader
+        # remote sampling loader: the seed set is a SERVER-side split
+        from glt_amd.sampler import RemoteSamplerInput
+
         opts = RemoteDistSamplingWorkerOptions(
             server_rank=0, num_workers=2, buffer_size="8MB",
             buffer_capacity=16, prefetch_size=2)
         loader = DistNeighborLoader(None, [2, 2],
-                                    input_nodes=torch.arange(VNUM),
+                                    input_nodes=RemoteSamplerInput(
+                                        split="train"),
                                     batch_size=5,
                                     worker_options=opts)
         for epoch in range(2):
