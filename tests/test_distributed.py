@@ -460,3 +460,53 @@ def test_dist_hetero_loader():
             p.terminate()
     for rank, err in results:
         assert err is None, f"rank {rank}:\n{err}"
+
+
+def _worker_subgraph_loader(rank, world, port, q):
+    try:
+        import torch
+
+        import glt_amd
+        from glt_amd.distributed import (CollocatedDistSamplingWorkerOptions,
+                                         DistSubGraphLoader, barrier,
+                                         init_worker_group)
+
+        glt_amd.seed_everything(3 + rank)
+        init_worker_group(world, rank)
+        ds = _build_partition(rank)
+        opts = CollocatedDistSamplingWorkerOptions(
+            master_addr="127.0.0.1", master_port=port)
+        seeds = torch.arange(rank, 12, 2)
+        loader = DistSubGraphLoader(ds, input_nodes=seeds, batch_size=6,
+                                    worker_options=opts)
+        for data in loader:
+            node = data.node
+            ei = data.edge_index
+            diff = (node[ei[1]] - node[ei[0]]) % VNUM
+            assert ((diff == 1) | (diff == 2)).all()
+        barrier()
+        q.put((rank, None))
+    except Exception:  # noqa: BLE001
+        import traceback
+
+        q.put((rank, traceback.format_exc()))
+
+
+@pytest.mark.timeout(180)
+def test_dist_subgraph_loader():
+    from glt_amd.utils import get_free_port
+
+    ctx = mp.get_context("spawn")
+    port = get_free_port()
+    q = ctx.Queue()
+    ps = [ctx.Process(target=_worker_subgraph_loader,
+                      args=(r, 2, port, q)) for r in range(2)]
+    for p in ps:
+        p.start()
+    results = [q.get(timeout=150) for _ in range(2)]
+    for p in ps:
+        p.join(timeout=30)
+        if p.is_alive():
+            p.terminate()
+    for rank, err in results:
+        assert err is None, f"rank {rank}:\n{err}"
