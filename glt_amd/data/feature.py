@@ -47,7 +47,8 @@ class Feature:
         (set when features were reordered or cached).
     """
 
-    def __init__(self, feature_tensor: torch.Tensor, split_ratio: float = 1.0,
+    def __init__(self, feature_tensor: torch.Tensor,
+                 split_ratio: Union[float, str] = 1.0,
                  device_group_list: Optional[List[DeviceGroup]] = None,
                  device: Optional[int] = None, with_gpu: bool = True,
                  dtype: Optional[torch.dtype] = None,
@@ -56,7 +57,11 @@ class Feature:
         if dtype is not None and feature_tensor.dtype != dtype:
             feature_tensor = feature_tensor.to(dtype)
         self.cpu_tensor = feature_tensor.contiguous()
-        self.split_ratio = float(split_ratio)
+        # split_ratio='auto': size the HBM-resident hot tier from free HBM
+        # at lazy-init time (288 GB/GPU usually fits everything; cap at 80%
+        # of free so training tensors keep headroom)
+        self._auto_split = split_ratio == "auto"
+        self.split_ratio = 1.0 if self._auto_split else float(split_ratio)
         self.device_group_list = device_group_list
         self.device = device
         self.with_gpu = with_gpu and torch.cuda.is_available()
@@ -77,6 +82,12 @@ class Feature:
         dev = self.device if self.device is not None else \
             torch.cuda.current_device()
         n = self.cpu_tensor.size(0)
+        if self._auto_split:
+            free_b, _ = torch.cuda.mem_get_info(dev)
+            row_bytes = self.cpu_tensor.size(1) * \
+                self.cpu_tensor.element_size()
+            self.split_ratio = min(
+                1.0, (free_b * 0.8) / max(row_bytes * n, 1))
         hot = int(n * min(max(self.split_ratio, 0.0), 1.0))
         store = _C.UnifiedFeatureStore(dev)
         groups = self.device_group_list

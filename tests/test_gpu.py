@@ -413,3 +413,12 @@ def test_gat_fused_matches_torch_path():
     assert torch.allclose(out, out2, atol=1e-4), (out - out2).abs().max()
     for a, b in zip(grads, grads2):
         assert torch.allclose(a, b, atol=1e-3), (a - b).abs().max()
+
+
+def test_feature_auto_split():
+    feats = torch.randn(1000, 32)
+    f = Feature(feats, split_ratio="auto", device=0, with_gpu=True)
+    f.lazy_init()
+    assert f.split_ratio == 1.0  # tiny matrix: fully HBM-resident
+    ids = torch.randint(0, 1000, (64,))
+    assert torch.equal(f[ids].cpu(), feats[ids])
