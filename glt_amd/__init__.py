@@ -13,6 +13,8 @@ import os
 
 # RCCL over xGMI: the host driver supports dmabuf IPC only.
 os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+# Quiet LibTorch warning spam (reference python/__init__.py:19 behavior).
+os.environ.setdefault("TORCH_CPP_LOG_LEVEL", "ERROR")
 
 import torch  # noqa: F401  (loads libc10/libtorch before our extension)
 
@@ -25,6 +27,7 @@ from .loader import (LinkLoader, LinkNeighborLoader, NeighborLoader,
 from .sampler import (EdgeSamplerInput, NegativeSampling, NeighborSampler,
                       NodeSamplerInput, RandomNegativeSampler,
                       SamplingConfig, SamplingType)
+from .typing import EdgeType, NodeType, as_str, reverse_edge_type
 from .utils import seed_everything
 
 __all__ = [
