@@ -189,7 +189,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("dim", &FeatureStorePy::dim);
 
   m.def("sage_gemm", &hip_sage_gemm, py::arg("A"), py::arg("B"),
-        py::arg("bias") = py::none());
+        py::arg("bias") = py::none(), py::arg("relu") = false);
   m.def("gat_fused_fwd", &hip_gat_fused_fwd);
   m.def("gat_fused_bwd", &hip_gat_fused_bwd);
   m.def("segment_mean_fwd", &hip_segment_mean_fwd);

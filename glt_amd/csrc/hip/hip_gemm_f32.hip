@@ -23,7 +23,7 @@ using f32x4 = __attribute__((__vector_size__(4 * sizeof(float)))) float;
 
 constexpr int BM = 64, BN = 64;
 
-template <int BK>
+template <int BK, bool RELU>
 __global__ __launch_bounds__(256)
 void sage_gemm_f32_kernel(const float* __restrict__ A,
                           const float* __restrict__ B,
@@ -124,6 +124,7 @@ void sage_gemm_f32_kernel(const float* __restrict__ A,
         if (row < M) {
           float v = acc[fi][fj][r];
           if (bias != nullptr) v += bias[col];
+          if (RELU) v = v > 0.f ? v : 0.f;
           C[row * N + col] = v;
         }
       }
@@ -141,6 +142,7 @@ using f32x16 = __attribute__((__vector_size__(16 * sizeof(float)))) float;
 
 constexpr int BM2 = 128, BN2 = 128, BK2 = 32;
 
+template <bool RELU>
 __global__ __launch_bounds__(256)
 void sage_gemm_f32_128_kernel(const float* __restrict__ A,
                               const float* __restrict__ B,
@@ -250,6 +252,7 @@ void sage_gemm_f32_128_kernel(const float* __restrict__ A,
         if (row < M) {
           float v = acc[fi][fj][reg];
           if (bias != nullptr) v += bias[col];
+          if (RELU) v = v > 0.f ? v : 0.f;
           C[row * N + col] = v;
         }
       }
@@ -261,7 +264,8 @@ void sage_gemm_f32_128_kernel(const float* __restrict__ A,
 
 // C = A @ B (+bias). A [M,K] row-major, B [K,N] row-major; N % 64 == 0.
 torch::Tensor hip_sage_gemm(const torch::Tensor& A, const torch::Tensor& B,
-                            const c10::optional<torch::Tensor>& bias) {
+                            const c10::optional<torch::Tensor>& bias,
+                            bool relu) {
   TORCH_CHECK(A.is_cuda() && B.is_cuda(), "device tensors required");
   TORCH_CHECK(A.scalar_type() == torch::kFloat32 &&
               B.scalar_type() == torch::kFloat32, "fp32 only");
@@ -278,25 +282,20 @@ torch::Tensor hip_sage_gemm(const torch::Tensor& A, const torch::Tensor& B,
   // staging; skinny-K projection shapes stay on the 64x64xBK64 kernel
   if (N % BN2 == 0 && M >= BM2 && K >= 384) {
     dim3 grid((uint32_t)((M + BM2 - 1) / BM2), (uint32_t)(N / BN2));
-    hipLaunchKernelGGL(sage_gemm_f32_128_kernel, grid, dim3(256), 0,
+    auto* kfn = relu ? sage_gemm_f32_128_kernel<true>
+                     : sage_gemm_f32_128_kernel<false>;
+    hipLaunchKernelGGL(kfn, grid, dim3(256), 0,
                        current_stream(), Ac.data_ptr<float>(),
                        Bc.data_ptr<float>(), bias_p, C.data_ptr<float>(),
                        M, K, N);
   } else {
     dim3 grid((uint32_t)((M + BM - 1) / BM), (uint32_t)(N / BN));
-    const char* bk_env = getenv("GLT_GEMM_BK");
-    const int bk = bk_env ? atoi(bk_env) : 16;
-    if (bk == 32) {
-      hipLaunchKernelGGL((sage_gemm_f32_kernel<32>), grid, dim3(256), 0,
-                         current_stream(), Ac.data_ptr<float>(),
-                         Bc.data_ptr<float>(), bias_p, C.data_ptr<float>(),
-                         M, K, N);
-    } else {
-      hipLaunchKernelGGL((sage_gemm_f32_kernel<16>), grid, dim3(256), 0,
-                         current_stream(), Ac.data_ptr<float>(),
-                         Bc.data_ptr<float>(), bias_p, C.data_ptr<float>(),
-                         M, K, N);
-    }
+    auto* kfn = relu ? sage_gemm_f32_kernel<16, true>
+                     : sage_gemm_f32_kernel<16, false>;
+    hipLaunchKernelGGL(kfn, grid, dim3(256), 0,
+                       current_stream(), Ac.data_ptr<float>(),
+                       Bc.data_ptr<float>(), bias_p, C.data_ptr<float>(),
+                       M, K, N);
   }
   return C;
 }

@@ -84,7 +84,8 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> hip_gat_fused_bwd(
 
 // --- f32 MFMA projection GEMM (hip_gemm_f32.hip) ----------------------------
 torch::Tensor hip_sage_gemm(const torch::Tensor& A, const torch::Tensor& B,
-                            const c10::optional<torch::Tensor>& bias);
+                            const c10::optional<torch::Tensor>& bias,
+                            bool relu);
 
 // --- memory plumbing (hip_mem.hip) -----------------------------------------
 // Device-dtype alias of (pinned/registered) host memory; keeps `src` alive.

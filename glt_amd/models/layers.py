@@ -33,9 +33,11 @@ class SAGEConv(nn.Module):
                       if root_weight else None)
 
     def forward(self, x, edge_index: torch.Tensor,
-                num_target: int = None, sorted_by_target: bool = True
-                ) -> torch.Tensor:
-        """x: [n, F] or (x_target, x_source) for bipartite relations."""
+                num_target: int = None, sorted_by_target: bool = True,
+                fuse_relu: bool = False) -> torch.Tensor:
+        """x: [n, F] or (x_target, x_source) for bipartite relations.
+        fuse_relu folds the activation into the projection GEMM epilogue
+        (MFMA path); the caller must then skip its own activation."""
         if isinstance(x, tuple):
             x_tgt, x_src = x
             n = num_target if num_target is not None else x_tgt.size(0)
@@ -72,10 +74,13 @@ class SAGEConv(nn.Module):
 
             w = torch.cat([self.lin_l.weight, self.lin_r.weight], dim=1)
             xin = torch.cat([agg, x[:n]], dim=1)
-            if use_mfma_linear(xin, w):
-                return mfma_linear(xin, w, self.lin_l.bias)
-            return F.linear(xin, w, self.lin_l.bias)
-        return self.lin_l(agg)
+            if use_mfma_linear(xin, w, relu=fuse_relu):
+                return mfma_linear(xin, w, self.lin_l.bias,
+                                   relu=fuse_relu)
+            out = F.linear(xin, w, self.lin_l.bias)
+            return F.relu(out) if fuse_relu else out
+        out = self.lin_l(agg)
+        return F.relu(out) if fuse_relu else out
 
 
 class GCNConv(nn.Module):

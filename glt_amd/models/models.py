@@ -49,15 +49,15 @@ class GraphSAGE(nn.Module):
         the seeds — ~(fan-out) x less compute on the deep layers."""
         L = len(self.convs)
         for i, conv in enumerate(self.convs):
+            act = i < L - 1  # fused into the projection GEMM epilogue
             trim = _layer_trim(num_sampled_nodes, num_sampled_edges, L, i)
             if trim is not None:
                 n_in, n_edges, n_out = trim
                 x = conv(x[:n_in], edge_index[:, :n_edges],
-                         num_target=n_out)
+                         num_target=n_out, fuse_relu=act)
             else:
-                x = conv(x, edge_index)
-            if i < len(self.convs) - 1:
-                x = F.relu(x)
+                x = conv(x, edge_index, fuse_relu=act)
+            if act:
                 x = F.dropout(x, p=self.dropout, training=self.training)
         return x
 

@@ -422,3 +422,25 @@ def test_feature_auto_split():
     assert f.split_ratio == 1.0  # tiny matrix: fully HBM-resident
     ids = torch.randint(0, 1000, (64,))
     assert torch.equal(f[ids].cpu(), feats[ids])
+
+
+def test_mfma_linear_fused_relu():
+    from glt_amd.ops import mfma_linear
+
+    torch.manual_seed(2)
+    x = torch.randn(500, 200, device="cuda", requires_grad=True)
+    w = torch.randn(256, 200, device="cuda", requires_grad=True)
+    b = torch.randn(256, device="cuda", requires_grad=True)
+    out = mfma_linear(x, w, b, relu=True)
+    ref = torch.relu(torch.nn.functional.linear(x.detach(), w.detach(),
+                                                b.detach()))
+    assert torch.allclose(out, ref, atol=1e-3, rtol=1e-4)
+    g = torch.randn_like(out)
+    out.backward(g)
+    xr = x.detach().clone().requires_grad_()
+    wr = w.detach().clone().requires_grad_()
+    br = b.detach().clone().requires_grad_()
+    torch.relu(torch.nn.functional.linear(xr, wr, br)).backward(g)
+    assert torch.allclose(x.grad, xr.grad, atol=1e-3, rtol=1e-4)
+    assert torch.allclose(w.grad, wr.grad, atol=1e-2, rtol=1e-4)
+    assert torch.allclose(b.grad, br.grad, atol=1e-2, rtol=1e-4)
