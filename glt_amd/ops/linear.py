@@ -59,5 +59,8 @@ def use_mfma_linear(x: torch.Tensor, weight: torch.Tensor,
             and weight.dtype == torch.float32
             and weight.size(0) % 64 == 0):
         return False
-    # un-fused: only where the custom kernel beats rocBLAS outright
-    return relu or weight.size(0) <= 128
+    if weight.size(0) <= 128:
+        return True   # ~2x rocBLAS at narrow outputs
+    # at n=256 the kernel TIES rocBLAS for k<=256 and loses at deep K
+    # (rocBLAS picks a K-split there); fuse the ReLU only where we tie
+    return relu and weight.size(1) <= 256
