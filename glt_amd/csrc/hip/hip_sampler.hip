@@ -99,7 +99,18 @@ __global__ void weighted_sample_kernel(
 #pragma unroll
     for (int s = kWave / 2; s > 0; s >>= 1) tot += __shfl_down(tot, s);
     tot = __shfl(tot, 0);
-    if (tot <= 0.f) continue;
+    if (tot <= 0.f) {
+      // degenerate weights: fall back to uniform draws so the output is
+      // never left uninitialized
+      for (int64_t j = lane; j < k; j += kWave) {
+        DRng rng(call_seed ^ (uint64_t)r * 0xD6E8FEB86659FD93ull ^
+                 (uint64_t)j * 0xA24BAED4963EE407ull);
+        const int64_t idx = (int64_t)rng.uniform((uint64_t)deg);
+        out_nbrs[off + j] = indices[base + idx];
+        if (WITH_EID) out_eids[off + j] = eids[base + idx];
+      }
+      continue;
+    }
     // Each lane serves draws lane, lane+64, ... by CDF scan.
     for (int64_t j = lane; j < k; j += kWave) {
       DRng rng(call_seed ^ (uint64_t)r * 0xD6E8FEB86659FD93ull ^

@@ -444,3 +444,14 @@ def test_mfma_linear_fused_relu():
     assert torch.allclose(x.grad, xr.grad, atol=1e-3, rtol=1e-4)
     assert torch.allclose(w.grad, wr.grad, atol=1e-2, rtol=1e-4)
     assert torch.allclose(b.grad, br.grad, atol=1e-2, rtol=1e-4)
+
+
+def test_weighted_sampler_zero_weights_gpu():
+    topo = Topology(torch.tensor([[0, 0, 0], [1, 2, 3]]),
+                    edge_weights=torch.zeros(3), num_nodes=4)
+    nbrs, num, _ = _C.sample_neighbors(
+        topo.indptr.cuda(), topo.indices.cuda(),
+        torch.zeros(50, dtype=torch.long, device="cuda"), 2,
+        edge_weights=topo.edge_weights.cuda(), weighted=True)
+    vals = nbrs.cpu()
+    assert ((vals >= 1) & (vals <= 3)).all()  # valid ids, no garbage
