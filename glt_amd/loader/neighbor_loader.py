@@ -31,9 +31,26 @@ class NeighborLoader(NodeLoader):
     def __next__(self):
         data = super().__next__()
         if self.as_pyg_v1:
+            # PyG-v1 contract: one EdgeIndex per layer, deepest hop first,
+            # sizes = (num rows of the layer input, num rows of its output)
             from ..sampler.base import EdgeIndex
 
-            n = data.node.numel()
-            return (data.batch_size, data.node,
-                    [EdgeIndex(data.edge_index, data.edge, (n, n))])
+            nsn = [int(v) for v in (data.num_sampled_nodes or [])]
+            nse = [int(v) for v in (data.num_sampled_edges or [])]
+            if nsn and nse:
+                adjs = []
+                n_pref = [sum(nsn[:i + 1]) for i in range(len(nsn))]
+                e_off = 0
+                for h, ne in enumerate(nse):
+                    ei = data.edge_index[:, e_off:e_off + ne]
+                    eid = (data.edge[e_off:e_off + ne]
+                           if data.edge is not None else None)
+                    size = (n_pref[min(h + 1, len(n_pref) - 1)], n_pref[h])
+                    adjs.append(EdgeIndex(ei, eid, size))
+                    e_off += ne
+                adjs = adjs[::-1]
+            else:
+                n = data.node.numel()
+                adjs = [EdgeIndex(data.edge_index, data.edge, (n, n))]
+            return data.batch_size, data.node, adjs
         return data

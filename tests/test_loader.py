@@ -45,11 +45,17 @@ def test_neighbor_loader_basic(ring_graph):
 
 def test_neighbor_loader_pyg_v1(ring_graph):
     ds = make_dataset(ring_graph)
-    loader = NeighborLoader(ds, [2], input_nodes=torch.arange(10),
+    loader = NeighborLoader(ds, [2, 2], input_nodes=torch.arange(10),
                             batch_size=10, as_pyg_v1=True, with_edge=True)
     bs, node, adjs = next(iter(loader))
     assert bs == 10
-    assert len(adjs) == 1
+    assert len(adjs) == 2  # one per hop, deepest first
+    # the LAST adj (hop 1) has target side = the seeds
+    assert adjs[-1].size[1] == 10
+    assert (adjs[-1].edge_index[0] < 10).all()
+    # deepest adj covers more rows than the shallow one
+    assert adjs[0].size[0] >= adjs[-1].size[0]
+    assert adjs[0].e_id is not None
 
 
 def test_link_neighbor_loader_binary(ring_graph):
