@@ -66,8 +66,19 @@ class DistNeighborSampler:
                  edge_dir: Literal["in", "out"] = "out",
                  device: Optional[torch.device] = None,
                  concurrency: int = 4,
-                 channel: Optional[ChannelBase] = None):
+                 channel: Optional[ChannelBase] = None,
+                 use_all2all: bool = False):
+        """use_all2all: collect remote feature rows with
+        torch.distributed.all_to_all_single (RCCL over xGMI intra-node)
+        instead of per-partition RPC pulls.  Requires one rank per
+        partition stepping in LOCKSTEP (same batch count, shuffle
+        deterministic) and forces concurrency=1 so the collectives stay
+        ordered (parity: the reference's optional gloo use_all2all path,
+        reference dist_sampling_producer.py:73-80)."""
         self.data = data
+        self.use_all2all = use_all2all
+        if use_all2all:
+            concurrency = 1
         self.num_neighbors = num_neighbors
         self.with_edge = with_edge
         self.with_weight = with_weight
@@ -402,6 +413,13 @@ class DistNeighborSampler:
                                      edge_attr=ea or None)
 
     async def _collect_one(self, loop, kind, ids, type_key):
+        if self.use_all2all:
+            import torch.distributed as dist
+
+            if dist.is_initialized() and \
+                    dist.get_world_size() == self.data.num_partitions:
+                return self.dist_feature.all2all_get(
+                    kind, ids.cpu(), type_key).to(self.sampler.device)
         futures, positions, local_vals, local_pos = \
             self.dist_feature.async_get(kind, ids, type_key)
         if not futures:

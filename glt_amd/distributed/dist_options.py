@@ -12,7 +12,9 @@ class _BasicDistSamplingWorkerOptions:
                  worker_concurrency: int = 4,
                  master_addr: Optional[str] = None,
                  master_port: Optional[Union[int, str]] = None,
-                 num_rpc_threads: int = 16, rpc_timeout: float = 180.0):
+                 num_rpc_threads: int = 16, rpc_timeout: float = 180.0,
+                 use_all2all: bool = False):
+        self.use_all2all = use_all2all
         self.num_workers = num_workers
         self.worker_devices = worker_devices
         self.worker_concurrency = min(max(worker_concurrency, 1), 32)
@@ -41,9 +43,10 @@ class CollocatedDistSamplingWorkerOptions(_BasicDistSamplingWorkerOptions):
     """Sample synchronously in the training process."""
 
     def __init__(self, master_addr=None, master_port=None,
-                 num_rpc_threads: int = 16, rpc_timeout: float = 180.0):
+                 num_rpc_threads: int = 16, rpc_timeout: float = 180.0,
+                 use_all2all: bool = False):
         super().__init__(1, None, 1, master_addr, master_port,
-                         num_rpc_threads, rpc_timeout)
+                         num_rpc_threads, rpc_timeout, use_all2all)
 
 
 class MpDistSamplingWorkerOptions(_BasicDistSamplingWorkerOptions):

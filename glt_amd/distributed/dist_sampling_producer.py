@@ -194,7 +194,9 @@ class DistCollocatedSamplingProducer:
             collect_features=self.config.collect_features,
             edge_dir=self.config.edge_dir, device=self.device,
             concurrency=self.worker_options.worker_concurrency,
-            channel=None)
+            channel=None,
+            use_all2all=getattr(self.worker_options, "use_all2all",
+                                False))
 
     def reset(self):
         n = len(self.seeds_input)

@@ -510,3 +510,61 @@ def test_dist_subgraph_loader():
             p.terminate()
     for rank, err in results:
         assert err is None, f"rank {rank}:\n{err}"
+
+
+def _worker_all2all_loader(rank, world, port, dist_port, q):
+    try:
+        import os
+
+        import torch
+        import torch.distributed as dist
+
+        import glt_amd
+        from glt_amd.distributed import (CollocatedDistSamplingWorkerOptions,
+                                         DistNeighborLoader, barrier,
+                                         init_worker_group)
+
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(dist_port)
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        glt_amd.seed_everything(21 + rank)
+        init_worker_group(world, rank)
+        ds = _build_partition(rank)
+        opts = CollocatedDistSamplingWorkerOptions(
+            master_addr="127.0.0.1", master_port=port, use_all2all=True)
+        seeds = torch.arange(rank, VNUM, 2)  # equal shards -> lockstep
+        loader = DistNeighborLoader(ds, [2, 2], input_nodes=seeds,
+                                    batch_size=5, shuffle=False,
+                                    worker_options=opts)
+        n = 0
+        for data in loader:
+            _check_batch(data)
+            n += 1
+        assert n == 4
+        barrier()
+        dist.destroy_process_group()
+        q.put((rank, None))
+    except Exception:  # noqa: BLE001
+        import traceback
+
+        q.put((rank, traceback.format_exc()))
+
+
+@pytest.mark.timeout(240)
+def test_dist_loader_all2all_features():
+    from glt_amd.utils import get_free_port
+
+    ctx = mp.get_context("spawn")
+    port, dport = get_free_port(), get_free_port()
+    q = ctx.Queue()
+    ps = [ctx.Process(target=_worker_all2all_loader,
+                      args=(r, 2, port, dport, q)) for r in range(2)]
+    for p in ps:
+        p.start()
+    results = [q.get(timeout=220) for _ in range(2)]
+    for p in ps:
+        p.join(timeout=30)
+        if p.is_alive():
+            p.terminate()
+    for rank, err in results:
+        assert err is None, f"rank {rank}:\n{err}"
