@@ -73,3 +73,28 @@ class TableDataset(Dataset):
     def _load_labels(self, table, num_threads, capacity):
         recs = self._read_all(table, capacity)
         self.init_node_labels(torch.tensor([int(r[1]) for r in recs]))
+
+
+class DistTableDataset(TableDataset):
+    """Distributed ODPS table ingestion (capability parity: reference
+    python/distributed/dist_table_dataset.py:149-353): every rank reads its
+    slice of the tables, then the chunks are exchanged to their owning
+    partitions with DistRandomPartitioner.  Requires `common_io`."""
+
+    def __init__(self, edge_table=None, node_table=None, label_table=None,
+                 num_nodes: int = 0, num_threads: int = 4,
+                 capacity: int = 1 << 16, **kwargs):
+        super().__init__(edge_table, node_table, label_table, num_threads,
+                         capacity, **kwargs)
+        from ..distributed.dist_random_partitioner import \
+            DistRandomPartitioner
+
+        rows, cols, _ = self.graph.topo.to_coo()
+        feats = (self.node_features.cpu_tensor
+                 if self.node_features is not None else None)
+        self._partitioner = DistRandomPartitioner(
+            num_nodes or self.graph.num_nodes,
+            torch.stack([rows, cols]), local_node_feat=feats)
+
+    def partition(self):
+        return self._partitioner.partition()

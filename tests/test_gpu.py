@@ -455,3 +455,41 @@ def test_weighted_sampler_zero_weights_gpu():
         edge_weights=topo.edge_weights.cuda(), weighted=True)
     vals = nbrs.cpu()
     assert ((vals >= 1) & (vals <= 3)).all()  # valid ids, no garbage
+
+
+def test_link_loader_gpu(ring_graph):
+    from glt_amd import Dataset, LinkNeighborLoader
+    from glt_amd.sampler import NegativeSampling
+
+    glt_amd.seed_everything(4)
+    ds = Dataset()
+    ds.init_graph(edge_index=ring_graph["edge_index"], graph_mode="CUDA",
+                  num_nodes=40, device=0)
+    ds.init_node_features(ring_graph["feats"], split_ratio=1.0, device=0)
+    loader = LinkNeighborLoader(
+        ds, [2], edge_label_index=ring_graph["edge_index"],
+        neg_sampling=NegativeSampling("binary"), batch_size=8,
+        device=torch.device("cuda", 0))
+    data = next(iter(loader))
+    assert data.edge_label_index.is_cuda
+    pos = data.edge_label_index[:, :8].cpu()
+    node = data.node.cpu()
+    diff = (node[pos[0]] - node[pos[1]]) % 40
+    assert ((diff == 1) | (diff == 2)).all()
+
+
+def test_subgraph_loader_gpu(ring_graph):
+    from glt_amd import Dataset
+    from glt_amd.loader import SubGraphLoader
+
+    ds = Dataset()
+    ds.init_graph(edge_index=ring_graph["edge_index"], graph_mode="CUDA",
+                  num_nodes=40, device=0)
+    ds.init_node_features(ring_graph["feats"], split_ratio=1.0, device=0)
+    loader = SubGraphLoader(ds, input_nodes=torch.arange(6), batch_size=6,
+                            device=torch.device("cuda", 0))
+    data = next(iter(loader))
+    got = set(zip(data.node[data.edge_index[0]].cpu().tolist(),
+                  data.node[data.edge_index[1]].cpu().tolist()))
+    expect = {(v, v + 1) for v in range(5)} | {(v, v + 2) for v in range(4)}
+    assert got == expect
