@@ -47,6 +47,8 @@ def parse_args():
     p.add_argument("--amp", action="store_true",
                    help="bf16 autocast for model math (default fp32, "
                         "matching the reference)")
+    p.add_argument("--compile", action="store_true",
+                   help="torch.compile the model (dynamic shapes)")
     p.add_argument("--feature-mode", type=str, default="replicated",
                    choices=["replicated", "xgmi-shard"],
                    help="xgmi-shard: features sharded across ranks' HBM, "
@@ -152,6 +154,8 @@ def main():
 
     model = GraphSAGE(args.feat_dim, args.hidden, len(fanout),
                       out_channels=args.classes).to(device)
+    if args.compile:
+        model = torch.compile(model, dynamic=True)
     if world > 1:
         model = torch.nn.parallel.DistributedDataParallel(
             model, device_ids=[local_rank] if has_gpu else None)
