@@ -49,6 +49,9 @@ def parse_args():
                         "matching the reference)")
     p.add_argument("--compile", action="store_true",
                    help="torch.compile the model (dynamic shapes)")
+    p.add_argument("--capture", action="store_true",
+                   help="hipGraph-capture fwd+bwd+opt on statically padded "
+                        "shapes (one replay per step)")
     p.add_argument("--feature-mode", type=str, default="replicated",
                    choices=["replicated", "xgmi-shard"],
                    help="xgmi-shard: features sharded across ranks' HBM, "
@@ -106,6 +109,109 @@ def build_synthetic(args, device, rank):
     return graph, feats, labels
 
 
+class _CapturedStep:
+    """hipGraph capture of forward+backward+Adam over statically padded
+    batch shapes (the step is launch/python-bound, not kernel-bound — see
+    profiles/; one graph.replay() runs the whole ~55-kernel step).
+
+    Static layout: region r holds hop-r nodes, sized (no-dedup cap) + 1
+    reserved always-pad row; hop-r edge segment is sized frontier_cap * k,
+    real edges first (targets remapped into the padded row space), then
+    pad edges targeting region (r-1)'s reserved row with src = 0.  That
+    keeps targets globally sorted for the fused segment kernels, keeps
+    pad outputs out of every real row, and makes every shape a constant of
+    (batch, fanout) so real batches always fit.
+    """
+
+    def __init__(self, model, opt, fanout, batch, feat_dim, classes,
+                 device):
+        self.device = device
+        caps = [batch]
+        for k in fanout:
+            caps.append(caps[-1] * k)
+        self.nsn = [c + 1 for c in caps]            # +1 reserved pad row
+        self.nse = [self.nsn[i] * fanout[i] for i in range(len(fanout))]
+        self.P = [0]
+        for v in self.nsn:
+            self.P.append(self.P[-1] + v)
+        self.n_cap = self.P[-1]
+        self.e_cap = sum(self.nse)
+        self.batch = batch
+        self.X = torch.zeros(self.n_cap, feat_dim, device=device)
+        self.EI = torch.empty(2, self.e_cap, dtype=torch.long,
+                              device=device)
+        self.Y = torch.zeros(batch, dtype=torch.long, device=device)
+        pad_tgt = torch.empty(self.e_cap, dtype=torch.long)
+        e_off = 0
+        for h, e in enumerate(self.nse):
+            pad_tgt[e_off:e_off + e] = self.P[h + 1] - 1  # reserved row
+            e_off += e
+        self.pad_tgt = pad_tgt.to(device)
+        self.model, self.opt = model, opt
+        self._graph = None
+        self._warmups = 0
+        self._loss = None
+
+    def _load(self, data):
+        nsn = [int(v) for v in data.num_sampled_nodes]
+        nse = [int(v) for v in data.num_sampled_edges]
+        pr = [0]
+        for v in nsn:
+            pr.append(pr[-1] + v)
+        # per-row shift: compact row space -> padded row space
+        shift = torch.zeros(pr[-1], dtype=torch.long, device=self.device)
+        for h in range(1, len(nsn)):
+            delta = self.P[h] - pr[h]
+            if delta:
+                shift[pr[h]:pr[h + 1]] = delta
+        # node features per region
+        for h in range(len(nsn)):
+            self.X[self.P[h]:self.P[h] + nsn[h]].copy_(
+                data.x[pr[h]:pr[h + 1]], non_blocking=True)
+        # edges: pads first, then real per static hop segment
+        self.EI[0].copy_(self.pad_tgt, non_blocking=True)
+        self.EI[1].fill_(0)
+        e_real = 0
+        e_static = 0
+        for h in range(len(nse)):
+            seg = data.edge_index[:, e_real:e_real + nse[h]]
+            self.EI[0, e_static:e_static + nse[h]].copy_(
+                seg[0] + shift[seg[0]])
+            self.EI[1, e_static:e_static + nse[h]].copy_(
+                seg[1] + shift[seg[1]])
+            e_real += nse[h]
+            e_static += self.nse[h]
+        self.Y.copy_(data.y[:self.batch], non_blocking=True)
+
+    def _run(self):
+        self.opt.zero_grad(set_to_none=False)
+        out = self.model(self.X, self.EI, self.nsn, self.nse)[:self.batch]
+        loss = F.cross_entropy(out, self.Y)
+        loss.backward()
+        self.opt.step()
+        return loss
+
+    def step(self, data):
+        self._load(data)
+        if self._graph is not None:
+            self._graph.replay()
+            return self._loss
+        self._warmups += 1
+        if self._warmups <= 3:
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                loss = self._run()
+            torch.cuda.current_stream().wait_stream(s)
+            return loss
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self._loss = self._run()
+        self._graph = g
+        self._graph.replay()
+        return self._loss
+
+
 def main():
     args = parse_args()
     rank = int(os.environ.get("RANK", 0))
@@ -159,14 +265,21 @@ def main():
     if world > 1:
         model = torch.nn.parallel.DistributedDataParallel(
             model, device_ids=[local_rank] if has_gpu else None)
-    opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3,
+                           capturable=bool(args.capture and has_gpu))
 
     seeds = torch.arange(args.nodes, device=device)
     loader = NeighborLoader(ds, fanout, input_nodes=seeds,
                             batch_size=args.batch_size, shuffle=True,
+                            drop_last=args.capture,
                             device=device, to_device=device,
                             prefetch=args.prefetch if has_gpu else 0)
     it = iter(loader)
+
+    captured = None
+    if args.capture and has_gpu and world == 1:
+        captured = _CapturedStep(model, opt, fanout, args.batch_size,
+                                 args.feat_dim, args.classes, device)
 
     import contextlib
 
@@ -182,6 +295,8 @@ def main():
         except StopIteration:
             it = iter(loader)
             data = next(it)
+        if captured is not None:
+            return captured.step(data)
         opt.zero_grad(set_to_none=True)
         with amp_ctx():
             out = model(data.x, data.edge_index, data.num_sampled_nodes,
