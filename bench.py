@@ -50,8 +50,11 @@ def parse_args():
     p.add_argument("--compile", action="store_true",
                    help="torch.compile the model (dynamic shapes)")
     p.add_argument("--capture", action="store_true",
-                   help="hipGraph-capture fwd+bwd+opt on statically padded "
-                        "shapes (one replay per step)")
+                   help="EXPERIMENTAL: hipGraph-capture fwd+bwd+opt on "
+                        "statically padded shapes. Correct but measured 6x "
+                        "slower on ROCm 7.0 (hipGraph per-node overhead "
+                        "~130us dominates this many-small-kernel step); "
+                        "kept as a recorded experiment, default off")
     p.add_argument("--feature-mode", type=str, default="replicated",
                    choices=["replicated", "xgmi-shard"],
                    help="xgmi-shard: features sharded across ranks' HBM, "

@@ -29,7 +29,8 @@ def gat_softmax_aggregate(h_src, a_src, a_dst, tgt, src, n_tgt,
                           negative_slope=0.2):
     """out[t,h,:] = sum_e softmax_t(leaky(a_dst[t,h]+a_src[src_e,h]))
     * h_src[src_e,h,:], over edges sorted by target."""
-    boundaries = torch.arange(n_tgt + 1, device=tgt.device)
-    offsets = torch.searchsorted(tgt, boundaries)
+    from .segment import _boundaries
+
+    offsets = torch.searchsorted(tgt, _boundaries(n_tgt, tgt.device))
     return _GatFused.apply(h_src, a_src, a_dst, src.contiguous(), offsets,
                            negative_slope)

@@ -6,6 +6,20 @@ index_add on CPU or unsorted input.
 """
 import torch
 
+_boundary_cache = {}
+
+
+def _boundaries(n: int, device) -> torch.Tensor:
+    """Cached 0..n arange per device (saves one kernel + one python op per
+    conv call; the step is launch-bound)."""
+    key = (device.type, device.index)
+    buf = _boundary_cache.get(key)
+    if buf is None or buf.numel() <= n:
+        buf = torch.arange(max(n + 1, 1 << 20), device=device)
+        _boundary_cache[key] = buf
+    return buf[: n + 1]
+
+
 
 class _SegmentMean(torch.autograd.Function):
     @staticmethod
@@ -32,7 +46,6 @@ def segment_mean(x: torch.Tensor, tgt: torch.Tensor, src: torch.Tensor,
     Requires tgt ascending (glt_amd batches satisfy this); returns
     [n_tgt, F].
     """
-    boundaries = torch.arange(n_tgt + 1, device=tgt.device)
-    offsets = torch.searchsorted(tgt, boundaries)
+    offsets = torch.searchsorted(tgt, _boundaries(n_tgt, tgt.device))
     return _SegmentMean.apply(x.contiguous(), src.contiguous(), offsets,
                               x.size(0))
