@@ -23,14 +23,20 @@ def _degree(target: torch.Tensor, num_nodes: int) -> torch.Tensor:
 
 
 class SAGEConv(nn.Module):
-    """GraphSAGE mean aggregator."""
+    """GraphSAGE mean aggregator.
+
+    The neighbor- and root-projections live in ONE [out, 2*in] parameter
+    applied to [agg | x]: a single GEMM per layer and no per-step weight
+    concatenation (the training step is launch-bound; see profiles/).
+    """
 
     def __init__(self, in_channels: int, out_channels: int,
                  root_weight: bool = True, bias: bool = True):
         super().__init__()
-        self.lin_l = nn.Linear(in_channels, out_channels, bias=bias)
-        self.lin_r = (nn.Linear(in_channels, out_channels, bias=False)
-                      if root_weight else None)
+        self.in_channels = in_channels
+        self.root_weight = root_weight
+        self.lin = nn.Linear(2 * in_channels if root_weight else
+                             in_channels, out_channels, bias=bias)
 
     def forward(self, x, edge_index: torch.Tensor,
                 num_target: int = None, sorted_by_target: bool = True,
@@ -51,10 +57,11 @@ class SAGEConv(nn.Module):
                 agg = x_src.new_zeros(n, x_src.size(1))
                 agg.index_add_(0, tgt, x_src.index_select(0, src))
                 agg = agg / _degree(tgt, n).unsqueeze(1).to(x_src.dtype)
-            out = self.lin_l(agg)
-            if self.lin_r is not None:
-                out = out + self.lin_r(x_tgt[:n])
-            return out
+            if self.root_weight:
+                xin = torch.cat([agg, x_tgt[:n]], dim=1)
+            else:
+                xin = agg
+            return self.lin(xin)
         n = num_target if num_target is not None else x.size(0)
         tgt, src = edge_index[0], edge_index[1]
         if (x.is_cuda and x.dtype == torch.float32 and sorted_by_target):
@@ -67,19 +74,13 @@ class SAGEConv(nn.Module):
             agg = x.new_zeros(n, x.size(1))
             agg.index_add_(0, tgt, x.index_select(0, src))
             agg = agg / _degree(tgt, n).unsqueeze(1).to(x.dtype)
-        if self.lin_r is not None:
-            # one [n, 2F] x [2F, out] GEMM instead of two skinny K=F GEMMs
-            # (halves GEMM launches; wgrad reduces once over n)
-            from ..ops import mfma_linear, use_mfma_linear
+        from ..ops import mfma_linear, use_mfma_linear
 
-            w = torch.cat([self.lin_l.weight, self.lin_r.weight], dim=1)
-            xin = torch.cat([agg, x[:n]], dim=1)
-            if use_mfma_linear(xin, w, relu=fuse_relu):
-                return mfma_linear(xin, w, self.lin_l.bias,
-                                   relu=fuse_relu)
-            out = F.linear(xin, w, self.lin_l.bias)
-            return F.relu(out) if fuse_relu else out
-        out = self.lin_l(agg)
+        xin = torch.cat([agg, x[:n]], dim=1) if self.root_weight else agg
+        if use_mfma_linear(xin, self.lin.weight, relu=fuse_relu):
+            return mfma_linear(xin, self.lin.weight, self.lin.bias,
+                               relu=fuse_relu)
+        out = self.lin(xin)
         return F.relu(out) if fuse_relu else out
 
 
