@@ -568,3 +568,62 @@ def test_dist_loader_all2all_features():
             p.terminate()
     for rank, err in results:
         assert err is None, f"rank {rank}:\n{err}"
+
+
+def _worker_link_loader(rank, world, port, q):
+    try:
+        import torch
+
+        import glt_amd
+        from glt_amd.distributed import (CollocatedDistSamplingWorkerOptions,
+                                         DistLinkNeighborLoader, barrier,
+                                         init_worker_group)
+
+        glt_amd.seed_everything(13 + rank)
+        init_worker_group(world, rank)
+        ds = _build_partition(rank)
+        opts = CollocatedDistSamplingWorkerOptions(
+            master_addr="127.0.0.1", master_port=port)
+        # seed edges owned by this partition (src parity)
+        rows = torch.arange(rank, VNUM, 2)
+        eli = torch.stack([rows, (rows + 1) % VNUM])
+        loader = DistLinkNeighborLoader(
+            ds, [2], edge_label_index=eli, neg_sampling="binary",
+            batch_size=5, worker_options=opts)
+        n = 0
+        for data in loader:
+            assert data.edge_label_index is not None
+            # positives decode back to (v, v+1) pairs
+            pos = data.edge_label_index[:, :5]
+            src = data.node[pos[1]]
+            dst = data.node[pos[0]]
+            assert ((dst - src) % VNUM == 1).all()
+            assert (data.x == data.node.float().unsqueeze(1)).all()
+            n += 1
+        assert n == 4, n
+        barrier()
+        q.put((rank, None))
+    except Exception:  # noqa: BLE001
+        import traceback
+
+        q.put((rank, traceback.format_exc()))
+
+
+@pytest.mark.timeout(180)
+def test_dist_link_neighbor_loader():
+    from glt_amd.utils import get_free_port
+
+    ctx = mp.get_context("spawn")
+    port = get_free_port()
+    q = ctx.Queue()
+    ps = [ctx.Process(target=_worker_link_loader, args=(r, 2, port, q))
+          for r in range(2)]
+    for p in ps:
+        p.start()
+    results = [q.get(timeout=150) for _ in range(2)]
+    for p in ps:
+        p.join(timeout=30)
+        if p.is_alive():
+            p.terminate()
+    for rank, err in results:
+        assert err is None, f"rank {rank}:\n{err}"
