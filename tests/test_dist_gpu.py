@@ -263,3 +263,86 @@ def test_xgmi_sharded_feature_world1():
     ids = torch.randint(0, 100, (37,))
     assert torch.equal(f[ids].cpu(), feats[ids])
     dist.destroy_process_group()
+
+
+def _worker_hetero_gpu(rank, world, port, q):
+    try:
+        import torch
+
+        import glt_amd
+        from glt_amd.data import Feature
+        from glt_amd.distributed import (CollocatedDistSamplingWorkerOptions,
+                                         DistDataset, DistNeighborLoader,
+                                         barrier, init_worker_group)
+        from glt_amd.partition import GLTPartitionBook
+
+        glt_amd.seed_everything(17 + rank)
+        torch.cuda.set_device(0)
+        init_worker_group(world, rank)
+        rows, cols = [], []
+        for u in range(rank, VNUM, 2):
+            rows += [u, u]
+            cols += [(u + 1) % VNUM, (u + 2) % VNUM]
+        et = ("user", "buys", "item")
+        ds = DistDataset(num_partitions=2, partition_idx=rank)
+        ds.init_graph(edge_index={et: torch.tensor([rows, cols])},
+                      graph_mode="CUDA", num_nodes=VNUM, device=0)
+        pb = GLTPartitionBook(torch.arange(VNUM) % 2)
+        ds.node_pb = {"user": pb, "item": pb}
+        feats = torch.arange(VNUM, dtype=torch.float32).unsqueeze(1)
+        local = torch.arange(rank, VNUM, 2)
+        id2index = torch.full((VNUM,), -1, dtype=torch.long)
+        id2index[local] = torch.arange(local.numel())
+        ds.node_features = {
+            "user": Feature(feats[local], split_ratio=1.0, device=0,
+                            with_gpu=True, id2index=id2index),
+            "item": Feature(feats[local] * 2.0, split_ratio=1.0, device=0,
+                            with_gpu=True, id2index=id2index),
+        }
+        ds._node_feat_pb = {"user": pb, "item": pb}
+        opts = CollocatedDistSamplingWorkerOptions(
+            master_addr="127.0.0.1", master_port=port)
+        seeds = torch.arange(rank, VNUM, 2)
+        loader = DistNeighborLoader(
+            ds, [2, 2], input_nodes=("user", seeds), batch_size=5,
+            to_device=torch.device("cuda", 0), worker_options=opts)
+        n = 0
+        for data in loader:
+            ei = data[et].edge_index.cpu()
+            users = data["user"].node.cpu()[ei[0]]
+            items = data["item"].node.cpu()[ei[1]]
+            diff = (items - users) % VNUM
+            assert ((diff == 1) | (diff == 2)).all()
+            assert data["user"].x.is_cuda
+            assert (data["user"].x.cpu() ==
+                    data["user"].node.cpu().float().unsqueeze(1)).all()
+            assert (data["item"].x.cpu() ==
+                    data["item"].node.cpu().float().unsqueeze(1) * 2).all()
+            n += 1
+        assert n == 4, n
+        barrier()
+        q.put((rank, None))
+    except Exception:  # noqa: BLE001
+        import traceback
+
+        q.put((rank, traceback.format_exc()))
+
+
+@pytest.mark.timeout(240)
+def test_dist_hetero_two_ranks_one_gpu():
+    from glt_amd.utils import get_free_port
+
+    ctx = mp.get_context("spawn")
+    port = get_free_port()
+    q = ctx.Queue()
+    ps = [ctx.Process(target=_worker_hetero_gpu, args=(r, 2, port, q))
+          for r in range(2)]
+    for p in ps:
+        p.start()
+    results = [q.get(timeout=220) for _ in range(2)]
+    for p in ps:
+        p.join(timeout=30)
+        if p.is_alive():
+            p.terminate()
+    for rank, err in results:
+        assert err is None, f"rank {rank}:\n{err}"
