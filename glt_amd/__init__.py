@@ -19,8 +19,8 @@ os.environ.setdefault("TORCH_CPP_LOG_LEVEL", "ERROR")
 import torch  # noqa: F401  (loads libc10/libtorch before our extension)
 
 from . import _C  # native core (built in-tree; fail loudly if missing)
-from . import channel, data, distributed, loader, models, partition, \
-    sampler, utils
+from . import channel, data, distributed, loader, models, parallel, \
+    partition, sampler, utils
 from .data import Dataset, DeviceGroup, Feature, Graph, Topology
 from .loader import (LinkLoader, LinkNeighborLoader, NeighborLoader,
                      NodeLoader, SubGraphLoader)
