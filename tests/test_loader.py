@@ -302,3 +302,41 @@ def test_sageconv_bipartite_matches_stacked():
     a = conv((x_tgt, x_src), ei)
     b = conv_bipartite(conv, x_tgt, x_src, ei)
     assert torch.allclose(a, b, atol=1e-5), (a - b).abs().max()
+
+
+def test_hetero_dict_fanout(ring_graph):
+    """Per-edge-type fan-out dicts (PyG-style num_neighbors={etype: [...]})."""
+    ei = ring_graph["edge_index"]
+    e1 = ("user", "buys", "item")
+    e2 = ("user", "views", "item")
+    ds = Dataset()
+    ds.init_graph(edge_index={e1: ei, e2: ei.flip(1)}, graph_mode="CPU",
+                  num_nodes=40)
+    loader = NeighborLoader(ds, {e1: [2], e2: [0]},
+                            input_nodes=("user", torch.arange(10)),
+                            batch_size=5)
+    data = next(iter(loader))
+    assert data[e1].edge_index is not None
+    assert data[e1].edge_index.size(1) > 0
+    # e2 fanout 0: no edges sampled for that relation
+    assert e2 not in data.edge_types or data[e2].edge_index is None \
+        or data[e2].edge_index.size(1) == 0
+
+
+def test_weighted_loader_e2e(ring_graph):
+    glt_amd.seed_everything(9)
+    ds = Dataset()
+    w = torch.ones(80)
+    # make the (v -> v+1) edges overwhelmingly likely
+    ds.init_graph(edge_index=ring_graph["edge_index"],
+                  edge_weights=torch.where(
+                      torch.arange(80) % 2 == 0, 1000.0, 0.001),
+                  graph_mode="CPU", num_nodes=40)
+    ds.init_node_features(ring_graph["feats"], with_gpu=False)
+    loader = NeighborLoader(ds, [1], input_nodes=torch.arange(40),
+                            batch_size=40, with_weight=True)
+    data = next(iter(loader))
+    diff = (data.node[data.edge_index[1]] -
+            data.node[data.edge_index[0]]) % 40
+    # heavy edges dominate: nearly every draw is the +1 neighbor
+    assert (diff == 1).float().mean() > 0.9
