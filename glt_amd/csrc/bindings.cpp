@@ -103,6 +103,20 @@ struct DeviceInducer {
   int64_t count() { return hip_inducer_count(p.get()); }
 };
 
+// Deferred-sync multi-hop sampler (opaque HIP-side implementation).
+struct DeferredSamplerPy {
+  std::shared_ptr<DeferredSampler> p;
+  DeferredSamplerPy(std::vector<int64_t> fanout, int64_t batch_cap,
+                    int64_t device, bool with_eid)
+      : p(deferred_sampler_create(std::move(fanout), batch_cap, device,
+                                  with_eid)) {}
+  auto run(const torch::Tensor& indptr, const torch::Tensor& indices,
+           const c10::optional<torch::Tensor>& edge_ids,
+           const torch::Tensor& seeds) {
+    return deferred_sampler_run(p.get(), indptr, indices, edge_ids, seeds);
+  }
+};
+
 struct FeatureStorePy {
   std::shared_ptr<UnifiedFeatureStore> p;
   explicit FeatureStorePy(int64_t device) : p(ufs_create(device)) {}
@@ -180,6 +194,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("lookup", &DeviceInducer::lookup)
       .def("insert", &DeviceInducer::insert)
       .def("count", &DeviceInducer::count);
+
+  py::class_<DeferredSamplerPy>(m, "DeferredSampler")
+      .def(py::init<std::vector<int64_t>, int64_t, int64_t, bool>(),
+           py::arg("fanout"), py::arg("batch_cap"), py::arg("device") = 0,
+           py::arg("with_eid") = false)
+      .def("run", &DeferredSamplerPy::run, py::arg("indptr"),
+           py::arg("indices"), py::arg("edge_ids"), py::arg("seeds"));
 
   py::class_<FeatureStorePy>(m, "UnifiedFeatureStore")
       .def(py::init<int64_t>(), py::arg("device"))

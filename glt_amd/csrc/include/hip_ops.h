@@ -47,6 +47,22 @@ torch::Tensor hip_inducer_lookup(HIPInducer* ind, const torch::Tensor& ids);
 torch::Tensor hip_inducer_insert(HIPInducer* ind, const torch::Tensor& ids);
 int64_t hip_inducer_count(HIPInducer* ind);
 
+// --- deferred-sync multi-hop sampler (hip_deferred.hip) ---------------------
+// Runs an entire L-hop batch (sample + dedup + relabel) with zero host
+// round-trips; all counts stay on device until the caller's single read of
+// the returned stats tensor [n_seed, n_new_1..L, total_e_1..L].
+class DeferredSampler;
+std::shared_ptr<DeferredSampler> deferred_sampler_create(
+    std::vector<int64_t> fanout, int64_t batch_cap, int64_t device,
+    bool with_eid);
+std::tuple<std::vector<torch::Tensor>, std::vector<torch::Tensor>,
+           std::vector<torch::Tensor>, std::vector<torch::Tensor>,
+           torch::Tensor>
+deferred_sampler_run(DeferredSampler* s, const torch::Tensor& indptr,
+                     const torch::Tensor& indices,
+                     const c10::optional<torch::Tensor>& edge_ids,
+                     const torch::Tensor& seeds);
+
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor,
            c10::optional<torch::Tensor>>
 hip_node_subgraph(const torch::Tensor& indptr, const torch::Tensor& indices,

@@ -109,6 +109,21 @@ class NeighborSampler(BaseSampler):
         self.device = torch.device(device) if device is not None \
             else self._sample_device
         self._cpu_mode = self._sample_device.type == "cpu"
+        # Deferred-sync fast path (GPU homo uniform sampling): the whole
+        # multi-hop batch runs with device-resident counts and ONE host
+        # sync at the end (see csrc/hip/hip_deferred.hip).  Capacity
+        # buffers are worst-case (batch x prod(fanout)); gate on a sane
+        # edge cap so exotic fan-outs fall back to the classic path.
+        self._deferred_pool = []
+        nn = self.num_neighbors
+        self.use_deferred = (
+            not self._cpu_mode and not self.is_hetero
+            and isinstance(nn, (list, tuple)) and len(nn) > 0
+            and all(isinstance(k, int) and k > 0 for k in nn)
+            and not (self.with_weight and g0.edge_weights is not None))
+        self._deferred_with_eid = (self.with_edge
+                                   and g0.edge_ids is not None
+                                   if not self.is_hetero else False)
 
     # ------------------------------------------------------------------
     def _make_inducer(self):
@@ -169,8 +184,55 @@ class NeighborSampler(BaseSampler):
             return self._sample_from_nodes(seeds,
                                            metadata={"input_type": None})
 
+    # -- deferred-sync fast path ---------------------------------------
+    def _acquire_deferred(self, bs: int):
+        with self._lock:
+            for i, (cap, ds) in enumerate(self._deferred_pool):
+                if cap >= bs:
+                    return self._deferred_pool.pop(i)
+        cap = max(1024, 1 << (bs - 1).bit_length())
+        ds = self._C.DeferredSampler(
+            list(self.num_neighbors), cap,
+            self._sample_device.index or 0, self._deferred_with_eid)
+        return (cap, ds)
+
+    def _release_deferred(self, entry):
+        with self._lock:
+            if len(self._deferred_pool) < 8:
+                self._deferred_pool.append(entry)
+
+    def _edge_cap(self, bs: int) -> int:
+        total, cap = 0, bs
+        for k in self.num_neighbors:
+            cap *= k
+            total += cap
+        return total
+
+    def _sample_from_nodes_deferred(self, seeds: torch.Tensor,
+                                    metadata=None) -> SamplerOutput:
+        g = self.graph
+        entry = self._acquire_deferred(seeds.numel())
+        node_parts, rows, cols, eids, stats = entry[1].run(
+            g.indptr, g.indices,
+            g.edge_ids if self._deferred_with_eid else None, seeds)
+        s = stats.cpu().tolist()  # the batch's single host sync
+        L = len(self.num_neighbors)
+        n_new, e_tot = s[:L + 1], s[L + 1:]
+        node = torch.cat([node_parts[i][:n_new[i]] for i in range(L + 1)])
+        row = torch.cat([rows[h][:e_tot[h]] for h in range(L)])
+        col = torch.cat([cols[h][:e_tot[h]] for h in range(L)])
+        edge = (torch.cat([eids[h][:e_tot[h]] for h in range(L)])
+                if eids else None)
+        self._release_deferred(entry)
+        return SamplerOutput(
+            node=node, row=row, col=col, edge=edge, batch=node[:n_new[0]],
+            num_sampled_nodes=n_new, num_sampled_edges=e_tot,
+            device=self.device, metadata=metadata)
+
     def _sample_from_nodes(self, seeds: torch.Tensor,
                            metadata=None) -> SamplerOutput:
+        if self.use_deferred and self._edge_cap(seeds.numel()) <= (1 << 25):
+            return self._sample_from_nodes_deferred(seeds, metadata)
         inducer = self._acquire_inducer()
         uniq_seeds = inducer.init_node(seeds)
         out_nodes = [uniq_seeds]

@@ -493,3 +493,83 @@ def test_subgraph_loader_gpu(ring_graph):
                   data.node[data.edge_index[1]].cpu().tolist()))
     expect = {(v, v + 1) for v in range(5)} | {(v, v + 2) for v in range(4)}
     assert got == expect
+
+
+def _ring_sampler(ring_graph, fanout, **kw):
+    from glt_amd.sampler import NeighborSampler
+
+    topo = Topology(ring_graph["edge_index"], num_nodes=40)
+    g = Graph(topo, mode="CUDA", device=0)
+    return NeighborSampler(g, fanout, **kw)
+
+
+def test_deferred_sampler_exact_match_full_fanout(ring_graph):
+    """Ring graph deg=2 with fanout [2, 2]: sampling is deterministic
+    (take-all), so the deferred-sync path must match the classic path
+    tensor-for-tensor."""
+    from glt_amd.sampler import NodeSamplerInput
+
+    s = _ring_sampler(ring_graph, [2, 2], with_edge=True)
+    assert s.use_deferred
+    seeds = torch.tensor([0, 5, 5, 17], device="cuda")
+    out_d = s.sample_from_nodes(NodeSamplerInput(node=seeds))
+    s.use_deferred = False
+    out_c = s.sample_from_nodes(NodeSamplerInput(node=seeds))
+    assert torch.equal(out_d.node, out_c.node)
+    assert torch.equal(out_d.row, out_c.row)
+    assert torch.equal(out_d.col, out_c.col)
+    assert torch.equal(out_d.edge, out_c.edge)
+    assert torch.equal(out_d.batch, out_c.batch)
+    assert out_d.num_sampled_nodes == out_c.num_sampled_nodes
+    assert out_d.num_sampled_edges == out_c.num_sampled_edges
+
+
+def test_deferred_sampler_structure_random_graph():
+    """Random graph, fanout < degree: every emitted edge must be a real
+    graph edge, node list unique & consistent with num_sampled_nodes."""
+    glt_amd.seed_everything(3)
+    n, deg = 5000, 20
+    src = torch.arange(n).repeat_interleave(deg)
+    dst = torch.randint(0, n, (n * deg,))
+    topo = Topology(torch.stack([src, dst]), num_nodes=n)
+    g = Graph(topo, mode="CUDA", device=0)
+    from glt_amd.sampler import NeighborSampler, NodeSamplerInput
+
+    s = NeighborSampler(g, [5, 3])
+    assert s.use_deferred
+    edge_set = set(zip(src.tolist(), dst.tolist()))
+    for trial in range(3):
+        seeds = torch.randint(0, n, (64,), device="cuda")
+        out = s.sample_from_nodes(NodeSamplerInput(node=seeds))
+        node = out.node.cpu()
+        assert node.numel() == len(set(node.tolist())), "dup nodes"
+        assert node.numel() == sum(out.num_sampled_nodes)
+        assert out.row.numel() == sum(out.num_sampled_edges)
+        gsrc = node[out.row.cpu()].tolist()
+        gdst = node[out.col.cpu()].tolist()
+        for a, b in zip(gsrc, gdst):
+            assert (a, b) in edge_set
+        # batch = unique seeds, first-occurrence order
+        seen, expect = set(), []
+        for v in seeds.cpu().tolist():
+            if v not in seen:
+                seen.add(v)
+                expect.append(v)
+        assert out.batch.cpu().tolist() == expect
+        # hop-1 rows reference seed-local ids only
+        e1 = out.num_sampled_edges[0]
+        assert out.row[:e1].max().item() < out.num_sampled_nodes[0]
+
+
+def test_deferred_sampler_capacity_growth(ring_graph):
+    """Batch bigger than the pooled capacity allocates a bigger instance."""
+    from glt_amd.sampler import NodeSamplerInput
+
+    from conftest import check_ring_edges
+
+    s = _ring_sampler(ring_graph, [2])
+    for bs in (8, 40, 33):
+        seeds = torch.randint(0, 40, (bs,), device="cuda")
+        out = s.sample_from_nodes(NodeSamplerInput(node=seeds))
+        check_ring_edges(out.node.cpu(),
+                         torch.stack([out.row.cpu(), out.col.cpu()]), 40)
