@@ -181,3 +181,19 @@ def test_cal_nbr_prob(ring_graph):
     assert abs(out[2].item() - 0.5) < 1e-5
     assert out[0].item() == 1.0  # seed stays
     assert out[3].item() == 0.0
+
+
+def test_deferred_gating(ring_graph):
+    """Deferred-sync path only engages for GPU homo uniform sampling with
+    plain positive fan-outs; everything else keeps the classic path."""
+    from glt_amd.data import Graph, Topology
+    from glt_amd.sampler import NeighborSampler
+
+    topo = Topology(ring_graph["edge_index"], num_nodes=40)
+    g = Graph(topo, mode="CPU")
+    assert not NeighborSampler(g, [2, 2]).use_deferred  # cpu mode
+    s = NeighborSampler(g, [-1])
+    assert not s.use_deferred  # full-neighbor fanout
+    # edge-capacity gate arithmetic
+    s2 = NeighborSampler(g, [15, 10, 5])
+    assert s2._edge_cap(1024) == 1024 * 15 * (1 + 10 + 10 * 5)
