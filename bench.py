@@ -268,8 +268,14 @@ def main():
     if world > 1:
         model = torch.nn.parallel.DistributedDataParallel(
             model, device_ids=[local_rank] if has_gpu else None)
-    opt = torch.optim.Adam(model.parameters(), lr=1e-3,
-                           capturable=bool(args.capture and has_gpu))
+    try:
+        # fused Adam: one multi-tensor HIP kernel per step (less python
+        # dispatch in the launch-bound regime); falls back to foreach.
+        opt = torch.optim.Adam(model.parameters(), lr=1e-3, fused=has_gpu,
+                               capturable=bool(args.capture and has_gpu))
+    except (RuntimeError, ValueError):
+        opt = torch.optim.Adam(model.parameters(), lr=1e-3,
+                               capturable=bool(args.capture and has_gpu))
 
     seeds = torch.arange(args.nodes, device=device)
     loader = NeighborLoader(ds, fanout, input_nodes=seeds,
