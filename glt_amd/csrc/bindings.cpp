@@ -215,6 +215,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gat_fused_bwd", &hip_gat_fused_bwd);
   m.def("segment_mean_fwd", &hip_segment_mean_fwd);
   m.def("segment_mean_bwd", &hip_segment_mean_bwd);
+  m.def("segment_mean_cat_fwd", &hip_segment_mean_cat_fwd);
+  m.def("segment_mean_cat_bwd", &hip_segment_mean_cat_bwd);
 
   // Memory plumbing
   m.def("host_mapped_view", &host_mapped_view, py::arg("src"),

@@ -62,6 +62,58 @@ __global__ void seg_mean_bwd_kernel(const float* __restrict__ dy,
   }
 }
 
+// Fused [mean-agg | x_root] assembly: out[t, :F] = segment mean,
+// out[t, F:] = x[t, :].  Saves the dim-1 torch.cat (two copyBuffer passes
+// over [n, F] per layer, ~260 us/step in the flagship profile) and its
+// launches; valid for the homo SAGE path where targets are the row prefix
+// of x (the sampler's local-id invariant).
+__global__ void seg_mean_cat_fwd_kernel(const float* __restrict__ x,
+                                        const int64_t* __restrict__ col,
+                                        const int64_t* __restrict__ off,
+                                        int64_t n_tgt, int64_t feat,
+                                        float* __restrict__ out) {
+  const int lane = threadIdx.x & (kWave - 1);
+  const int64_t wave =
+      (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) / kWave;
+  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) / kWave;
+  const int64_t ostride = 2 * feat;
+  for (int64_t t = wave; t < n_tgt; t += n_waves) {
+    const int64_t s = off[t], e = off[t + 1];
+    const float inv = e > s ? 1.0f / (float)(e - s) : 0.0f;
+    for (int64_t f = lane; f < feat; f += kWave) {
+      float acc = 0.f;
+      for (int64_t k = s; k < e; ++k) acc += x[col[k] * feat + f];
+      out[t * ostride + f] = acc * inv;
+      out[t * ostride + feat + f] = x[t * feat + f];
+    }
+  }
+}
+
+// dx[t, :] += dy[t, F:];  dx[col[e], :] += dy[t, :F] / deg(t)
+__global__ void seg_mean_cat_bwd_kernel(const float* __restrict__ dy,
+                                        const int64_t* __restrict__ col,
+                                        const int64_t* __restrict__ off,
+                                        int64_t n_tgt, int64_t feat,
+                                        float* __restrict__ dx) {
+  const int lane = threadIdx.x & (kWave - 1);
+  const int64_t wave =
+      (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) / kWave;
+  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) / kWave;
+  const int64_t ostride = 2 * feat;
+  for (int64_t t = wave; t < n_tgt; t += n_waves) {
+    const int64_t s = off[t], e = off[t + 1];
+    for (int64_t f = lane; f < feat; f += kWave)
+      atomicAdd(&dx[t * feat + f], dy[t * ostride + feat + f]);
+    if (e <= s) continue;
+    const float inv = 1.0f / (float)(e - s);
+    for (int64_t k = s; k < e; ++k) {
+      const int64_t c = col[k];
+      for (int64_t f = lane; f < feat; f += kWave)
+        atomicAdd(&dx[c * feat + f], dy[t * ostride + f] * inv);
+    }
+  }
+}
+
 int wave_grid(int64_t rows) {
   const int64_t waves = std::min<int64_t>(rows, (int64_t)kMaxBlocks * 4);
   return (int)std::min<int64_t>((waves * kWave + kBlock - 1) / kBlock,
@@ -87,6 +139,45 @@ torch::Tensor hip_segment_mean_fwd(const torch::Tensor& x,
                        out.data_ptr<float>());
   }
   return out;
+}
+
+torch::Tensor hip_segment_mean_cat_fwd(const torch::Tensor& x,
+                                       const torch::Tensor& col,
+                                       const torch::Tensor& offsets,
+                                       int64_t n_tgt) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kFloat32 &&
+                  x.is_contiguous(),
+              "segment_mean_cat: x must be contiguous fp32 on device");
+  TORCH_CHECK(x.size(0) >= n_tgt,
+              "segment_mean_cat: targets must be a row prefix of x");
+  const int64_t feat = x.size(1);
+  auto out = torch::empty({n_tgt, 2 * feat}, x.options());
+  if (n_tgt > 0) {
+    hipLaunchKernelGGL(seg_mean_cat_fwd_kernel, dim3(wave_grid(n_tgt)),
+                       dim3(kBlock), 0, current_stream(),
+                       x.data_ptr<float>(), col.data_ptr<int64_t>(),
+                       offsets.data_ptr<int64_t>(), n_tgt, feat,
+                       out.data_ptr<float>());
+  }
+  return out;
+}
+
+torch::Tensor hip_segment_mean_cat_bwd(const torch::Tensor& dy,
+                                       const torch::Tensor& col,
+                                       const torch::Tensor& offsets,
+                                       int64_t n_src) {
+  const int64_t n_tgt = dy.size(0);
+  const int64_t feat = dy.size(1) / 2;
+  auto dx = torch::zeros({n_src, feat}, dy.options());
+  if (n_tgt > 0) {
+    hipLaunchKernelGGL(seg_mean_cat_bwd_kernel, dim3(wave_grid(n_tgt)),
+                       dim3(kBlock), 0, current_stream(),
+                       dy.contiguous().data_ptr<float>(),
+                       col.data_ptr<int64_t>(),
+                       offsets.data_ptr<int64_t>(), n_tgt, feat,
+                       dx.data_ptr<float>());
+  }
+  return dx;
 }
 
 torch::Tensor hip_segment_mean_bwd(const torch::Tensor& dy,

@@ -85,6 +85,15 @@ torch::Tensor hip_segment_mean_bwd(const torch::Tensor& dy,
                                    const torch::Tensor& col,
                                    const torch::Tensor& offsets,
                                    int64_t n_src);
+// Fused [mean-agg | x-prefix] assembly (skips the dim-1 torch.cat).
+torch::Tensor hip_segment_mean_cat_fwd(const torch::Tensor& x,
+                                       const torch::Tensor& col,
+                                       const torch::Tensor& offsets,
+                                       int64_t n_tgt);
+torch::Tensor hip_segment_mean_cat_bwd(const torch::Tensor& dy,
+                                       const torch::Tensor& col,
+                                       const torch::Tensor& offsets,
+                                       int64_t n_src);
 
 // --- fused GAT edge softmax + aggregation (hip_gat.hip) ---------------------
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> hip_gat_fused_fwd(

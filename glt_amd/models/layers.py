@@ -64,19 +64,26 @@ class SAGEConv(nn.Module):
             return self.lin(xin)
         n = num_target if num_target is not None else x.size(0)
         tgt, src = edge_index[0], edge_index[1]
+        from ..ops import mfma_linear, use_mfma_linear
+
         if (x.is_cuda and x.dtype == torch.float32 and sorted_by_target):
             # fused wave-per-row segment mean (glt_amd batches are sorted
-            # by target local id by construction)
-            from ..ops import segment_mean
+            # by target local id by construction); with root_weight the
+            # kernel also assembles [agg | x[:n]] in place of a dim-1 cat
+            if self.root_weight:
+                from ..ops import segment_mean_cat
 
-            agg = segment_mean(x, tgt, src, n)
+                xin = segment_mean_cat(x, tgt, src, n)
+            else:
+                from ..ops import segment_mean
+
+                xin = segment_mean(x, tgt, src, n)
         else:
             agg = x.new_zeros(n, x.size(1))
             agg.index_add_(0, tgt, x.index_select(0, src))
             agg = agg / _degree(tgt, n).unsqueeze(1).to(x.dtype)
-        from ..ops import mfma_linear, use_mfma_linear
-
-        xin = torch.cat([agg, x[:n]], dim=1) if self.root_weight else agg
+            xin = torch.cat([agg, x[:n]], dim=1) if self.root_weight \
+                else agg
         if use_mfma_linear(xin, self.lin.weight, relu=fuse_relu):
             return mfma_linear(xin, self.lin.weight, self.lin.bias,
                                relu=fuse_relu)

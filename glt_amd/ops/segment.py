@@ -39,6 +39,30 @@ class _SegmentMean(torch.autograd.Function):
         return dx, None, None, None
 
 
+class _SegmentMeanCat(torch.autograd.Function):
+    """Fused ``[segment_mean(x) | x[:n_tgt]]`` — one kernel instead of
+    segment-mean + dim-1 torch.cat (which costs two full copyBuffer
+    passes over [n, F] per conv layer; see profiles/).  Homo-path only:
+    targets must be the row prefix of x."""
+
+    @staticmethod
+    def forward(ctx, x, col, offsets, n_tgt):
+        from .. import _C
+
+        ctx.save_for_backward(col, offsets)
+        ctx.n_src = x.size(0)
+        return _C.segment_mean_cat_fwd(x, col, offsets, n_tgt)
+
+    @staticmethod
+    def backward(ctx, dy):
+        from .. import _C
+
+        col, offsets = ctx.saved_tensors
+        dx = _C.segment_mean_cat_bwd(dy.contiguous(), col, offsets,
+                                     ctx.n_src)
+        return dx, None, None, None
+
+
 def segment_mean(x: torch.Tensor, tgt: torch.Tensor, src: torch.Tensor,
                  n_tgt: int) -> torch.Tensor:
     """Mean of x[src[e]] over contiguous tgt segments.
@@ -49,3 +73,12 @@ def segment_mean(x: torch.Tensor, tgt: torch.Tensor, src: torch.Tensor,
     offsets = torch.searchsorted(tgt, _boundaries(n_tgt, tgt.device))
     return _SegmentMean.apply(x.contiguous(), src.contiguous(), offsets,
                               x.size(0))
+
+
+def segment_mean_cat(x: torch.Tensor, tgt: torch.Tensor, src: torch.Tensor,
+                     n_tgt: int) -> torch.Tensor:
+    """[segment_mean | root] fused: returns [n_tgt, 2F] where the left F
+    columns are the neighbor mean and the right F columns are x[:n_tgt]."""
+    offsets = torch.searchsorted(tgt, _boundaries(n_tgt, tgt.device))
+    return _SegmentMeanCat.apply(x.contiguous(), src.contiguous(), offsets,
+                                 n_tgt)
