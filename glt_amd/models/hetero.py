@@ -21,7 +21,7 @@ class HeteroConv(nn.Module):
 
     def forward(self, x_dict: Dict[NodeType, torch.Tensor],
                 edge_index_dict: Dict[EdgeType, torch.Tensor]):
-        out: Dict[NodeType, List[torch.Tensor]] = {}
+        rels = []
         for etype, ei in edge_index_dict.items():
             key = "__".join(etype)
             if key not in self.convs:
@@ -30,21 +30,66 @@ class HeteroConv(nn.Module):
             # sampler emits edges keyed so that edge_index[0] is the
             # walked-from (seed-side) type.
             src_t, _, dst_t = etype
-            tgt_t = src_t  # seed side
-            x_tgt = x_dict.get(tgt_t)
-            x_src = x_dict.get(dst_t)
-            if x_tgt is None or x_src is None:
+            if x_dict.get(src_t) is None or x_dict.get(dst_t) is None:
                 continue
-            conv = self.convs[key]
-            try:
-                h = conv((x_tgt, x_src), ei)
-            except TypeError:
-                h = conv_bipartite(conv, x_tgt, x_src, ei)
-            out.setdefault(tgt_t, []).append(h)
+            rels.append((etype, key, ei))
+        if rels and all(isinstance(self.convs[k], GATConv)
+                        for _, k, _ in rels):
+            out = self._batched_gat(x_dict, rels)
+        else:
+            out = {}
+            for etype, key, ei in rels:
+                src_t, _, dst_t = etype
+                x_tgt, x_src = x_dict[src_t], x_dict[dst_t]
+                conv = self.convs[key]
+                try:
+                    h = conv((x_tgt, x_src), ei)
+                except TypeError:
+                    h = conv_bipartite(conv, x_tgt, x_src, ei)
+                out.setdefault(src_t, []).append(h)
         result = {}
         for t, hs in out.items():
             result[t] = torch.stack(hs).sum(0) if len(hs) > 1 else hs[0]
         return result
+
+    def _batched_gat(self, x_dict, rels):
+        """All-GAT layers: batch the per-relation projections of each node
+        type into ONE GEMM (RGAT spends most of its host+GEMM time on many
+        small per-relation Linears otherwise), then run each relation's
+        attention over views of the stacked output."""
+        needs: Dict[NodeType, List[str]] = {}
+        for etype, key, _ in rels:
+            for t in (etype[0], etype[2]):
+                keys = needs.setdefault(t, [])
+                if key not in keys:
+                    keys.append(key)
+        H = {}
+        for t, keys in needs.items():
+            x = x_dict[t]
+            convs = [self.convs[k] for k in keys]
+            in_dims = {c.lin.weight.size(1) for c in convs}
+            if len(keys) == 1 or len(in_dims) > 1:
+                for k, c in zip(keys, convs):
+                    H[(t, k)] = c.lin(x).view(x.size(0), c.heads,
+                                              c.out_channels)
+                continue
+            W = torch.cat([c.lin.weight for c in convs], dim=0)
+            h = torch.nn.functional.linear(x, W)
+            off = 0
+            for k, c in zip(keys, convs):
+                hc = c.heads * c.out_channels
+                # last-dim split of a stride-1 slice: a true view, no copy
+                H[(t, k)] = h[:, off:off + hc].view(
+                    x.size(0), c.heads, c.out_channels)
+                off += hc
+        out: Dict[NodeType, List[torch.Tensor]] = {}
+        for etype, key, ei in rels:
+            src_t, _, dst_t = etype
+            conv = self.convs[key]
+            h = conv.attend(H[(src_t, key)], H[(dst_t, key)], ei,
+                            x_dict[src_t].size(0))
+            out.setdefault(src_t, []).append(h)
+        return out
 
 
 def conv_bipartite(conv, x_tgt, x_src, edge_index):

@@ -149,6 +149,12 @@ class GATConv(nn.Module):
             nt = num_target if num_target is not None else x.size(0)
             h_tgt = h_src = self.lin(x).view(x.size(0), self.heads,
                                              self.out_channels)
+        return self.attend(h_tgt, h_src, edge_index, nt)
+
+    def attend(self, h_tgt, h_src, edge_index, nt):
+        """Attention + aggregation over pre-projected features
+        [n, heads, C] (lets HeteroConv batch the projections of all
+        relations sharing a node type into one GEMM)."""
         tgt, src = edge_index[0], edge_index[1]
         h = h_src
         alpha_src = (h_src * self.att_src).sum(-1)

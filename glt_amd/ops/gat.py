@@ -7,9 +7,11 @@ class _GatFused(torch.autograd.Function):
     def forward(ctx, h_src, a_src, a_dst, src, offsets, slope):
         from .. import _C
 
-        out, m, z = _C.gat_fused_fwd(h_src.contiguous(),
-                                     a_src.contiguous(),
-                                     a_dst.contiguous(), src, offsets,
+        # contiguity once, shared by fwd and the saved tensors bwd reads
+        h_src = h_src.contiguous()
+        a_src = a_src.contiguous()
+        a_dst = a_dst.contiguous()
+        out, m, z = _C.gat_fused_fwd(h_src, a_src, a_dst, src, offsets,
                                      slope)
         ctx.save_for_backward(h_src, a_src, a_dst, src, offsets, out, m, z)
         ctx.slope = slope
