@@ -27,6 +27,10 @@ def main():
     ap.add_argument("--steps", type=int, default=60)
     ap.add_argument("--warmup", type=int, default=10)
     ap.add_argument("--prefetch", type=int, default=3)
+    # pipeline diagnosis: time only the producer (loader) or only the
+    # consumer (model step on one cached batch)
+    ap.add_argument("--sampler-only", action="store_true")
+    ap.add_argument("--model-only", action="store_true")
     args = ap.parse_args()
 
     import glt_amd
@@ -81,13 +85,20 @@ def main():
     opt = torch.optim.Adam(model.parameters(), lr=1e-3)
     it = iter(loader)
 
+    cached = [next(it)] if args.model_only else None
+
     def step():
         nonlocal it
-        try:
-            data = next(it)
-        except StopIteration:
-            it = iter(loader)
-            data = next(it)
+        if cached is not None:
+            data = cached[0]
+        else:
+            try:
+                data = next(it)
+            except StopIteration:
+                it = iter(loader)
+                data = next(it)
+        if args.sampler_only:
+            return
         opt.zero_grad(set_to_none=True)
         out = model(data.x_dict, data.edge_index_dict,
                     predict_type="paper")
