@@ -82,7 +82,10 @@ def main():
     model = RGNN(list(ds.graph.keys()), dim, args.hidden, args.classes,
                  num_layers=len(fanout), n_heads=args.heads,
                  model=args.model).to(device)
-    opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+    try:
+        opt = torch.optim.Adam(model.parameters(), lr=1e-3, fused=has_gpu)
+    except (RuntimeError, ValueError):
+        opt = torch.optim.Adam(model.parameters(), lr=1e-3)
     it = iter(loader)
 
     cached = [next(it)] if args.model_only else None

@@ -20,7 +20,11 @@ class HeteroConv(nn.Module):
         self.aggr = aggr
 
     def forward(self, x_dict: Dict[NodeType, torch.Tensor],
-                edge_index_dict: Dict[EdgeType, torch.Tensor]):
+                edge_index_dict: Dict[EdgeType, torch.Tensor],
+                self_lins: Optional[Dict[NodeType, nn.Module]] = None):
+        """With `self_lins` (a per-node-type Linear dict, e.g. RGNN's self
+        projections), returns (conv_out, self_out); the self projections
+        ride along in the batched per-type GEMM where shapes allow."""
         rels = []
         for etype, ei in edge_index_dict.items():
             key = "__".join(etype)
@@ -33,9 +37,10 @@ class HeteroConv(nn.Module):
             if x_dict.get(src_t) is None or x_dict.get(dst_t) is None:
                 continue
             rels.append((etype, key, ei))
+        self_out = {} if self_lins is not None else None
         if rels and all(isinstance(self.convs[k], GATConv)
                         for _, k, _ in rels):
-            out = self._batched_gat(x_dict, rels)
+            out = self._batched_gat(x_dict, rels, self_lins, self_out)
         else:
             out = {}
             for etype, key, ei in rels:
@@ -47,16 +52,22 @@ class HeteroConv(nn.Module):
                 except TypeError:
                     h = conv_bipartite(conv, x_tgt, x_src, ei)
                 out.setdefault(src_t, []).append(h)
+        if self_out is not None:
+            for t, x in x_dict.items():
+                if t not in self_out and t in self_lins:
+                    self_out[t] = self_lins[t](x)
         result = {}
         for t, hs in out.items():
             result[t] = torch.stack(hs).sum(0) if len(hs) > 1 else hs[0]
-        return result
+        return result if self_out is None else (result, self_out)
 
-    def _batched_gat(self, x_dict, rels):
+    def _batched_gat(self, x_dict, rels, self_lins=None, self_out=None):
         """All-GAT layers: batch the per-relation projections of each node
         type into ONE GEMM (RGAT spends most of its host+GEMM time on many
         small per-relation Linears otherwise), then run each relation's
-        attention over views of the stacked output."""
+        attention over views of the stacked output.  When a matching
+        per-type self projection is supplied, its weight rides in the same
+        GEMM (bias added on the slice)."""
         needs: Dict[NodeType, List[str]] = {}
         for etype, key, _ in rels:
             for t in (etype[0], etype[2]):
@@ -68,12 +79,18 @@ class HeteroConv(nn.Module):
             x = x_dict[t]
             convs = [self.convs[k] for k in keys]
             in_dims = {c.lin.weight.size(1) for c in convs}
-            if len(keys) == 1 or len(in_dims) > 1:
+            if len(in_dims) > 1:
                 for k, c in zip(keys, convs):
                     H[(t, k)] = c.lin(x).view(x.size(0), c.heads,
                                               c.out_channels)
                 continue
-            W = torch.cat([c.lin.weight for c in convs], dim=0)
+            weights = [c.lin.weight for c in convs]
+            fold_self = (self_lins is not None and t in self_lins
+                         and self_lins[t].weight.size(1) == x.size(1))
+            if fold_self:
+                weights.append(self_lins[t].weight)
+            W = weights[0] if len(weights) == 1 \
+                else torch.cat(weights, dim=0)
             h = torch.nn.functional.linear(x, W)
             off = 0
             for k, c in zip(keys, convs):
@@ -82,6 +99,10 @@ class HeteroConv(nn.Module):
                 H[(t, k)] = h[:, off:off + hc].view(
                     x.size(0), c.heads, c.out_channels)
                 off += hc
+            if fold_self:
+                s = h[:, off:]
+                b = self_lins[t].bias
+                self_out[t] = s if b is None else s + b
         out: Dict[NodeType, List[torch.Tensor]] = {}
         for etype, key, ei in rels:
             src_t, _, dst_t = etype
@@ -143,18 +164,22 @@ class RGNN(nn.Module):
                 predict_type: Optional[NodeType] = None):
         h = x_dict
         for i, layer in enumerate(self.layers):
-            h_conv = layer(h, edge_index_dict)
+            h_conv, self_out = layer(h, edge_index_dict,
+                                     self_lins=self.self_lins[i])
             h_next = {}
             for t, v in h.items():
                 if t not in self.self_lins[i]:
                     continue
-                out = self.self_lins[i][t](v)
+                out = self_out[t]
                 hc = h_conv.get(t)
                 if hc is not None:
                     # conv output may cover fewer rows than the projection
                     # (message targets only); add on the common prefix
                     m = min(hc.size(0), out.size(0))
-                    out = torch.cat([out[:m] + hc[:m], out[m:]], dim=0)
+                    if m == out.size(0) and m == hc.size(0):
+                        out = out + hc  # full cover: skip the cat copy
+                    else:
+                        out = torch.cat([out[:m] + hc[:m], out[m:]], dim=0)
                 h_next[t] = F.dropout(F.relu(out), p=self.dropout,
                                       training=self.training)
             h = h_next

@@ -377,3 +377,26 @@ def test_hetero_batched_gat_matches_per_relation():
     for et in ets:
         assert convs[et].lin.weight.grad is not None
         assert convs[et].att_src.grad is not None
+
+
+def test_hetero_conv_folded_self_projection():
+    """self_lins riding in the batched GEMM must equal plain Linear."""
+    import torch
+    from glt_amd.models import GATConv
+    from glt_amd.models.hetero import HeteroConv
+
+    torch.manual_seed(1)
+    ets = [("u", "a", "v"), ("v", "b", "u")]
+    convs = {et: GATConv(8, 4, heads=2) for et in ets}
+    hc = HeteroConv(convs)
+    lins = torch.nn.ModuleDict(
+        {"u": torch.nn.Linear(8, 8), "v": torch.nn.Linear(8, 8),
+         "w": torch.nn.Linear(8, 8)})
+    x = {"u": torch.randn(6, 8), "v": torch.randn(5, 8),
+         "w": torch.randn(3, 8)}  # w: no relations, plain path
+    ei = {et: torch.stack([torch.sort(torch.randint(0, 5, (9,)))[0],
+                           torch.randint(0, 5, (9,))]) for et in ets}
+    out, self_out = hc(x, ei, self_lins=lins)
+    for t in ("u", "v", "w"):
+        assert torch.allclose(self_out[t], lins[t](x[t]), atol=1e-6), t
+    assert set(out) == {"u", "v"}
