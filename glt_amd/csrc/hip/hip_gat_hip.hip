@@ -1,23 +1,29 @@
 #include "hip/hip_runtime.h"
 /* Fused GAT edge-softmax + weighted aggregation (gfx950).
  *
- * Replaces the 6-kernel torch chain per GATConv (gather-add, leaky_relu,
- * scatter-amax, exp, scatter-sum, weighted index_add) and its sort-heavy
- * backward with one forward and one backward kernel.
+ * Replaces the whole torch chain per GATConv — including the per-node
+ * attention-logit precompute (h*att).sum(-1) — with one forward and one
+ * backward kernel.  The logits are recomputed per edge from rows the
+ * aggregation loads anyway (a 6-step wave reduction on data already in
+ * registers), which deletes ~8 small launches per relation per layer in
+ * RGAT's launch-bound regime.
  *
  * Layout: edges sorted by target (glt_amd batch invariant); one wave per
  * (target, head), lanes over the channel dim; the softmax runs online
  * (running max / rescaled sum) along the target's edge segment, so scores
  * are never materialized.
  *
- *   s_e   = leaky_relu(alpha_dst[t,h] + alpha_src[src_e,h])
+ *   s_e   = leaky_relu(<h_tgt[t,h,:], att_dst[h,:]> +
+ *                      <h_src[src_e,h,:], att_src[h,:]>)
  *   p_e   = exp(s_e - m_t) / Z_t
  *   out_t = sum_e p_e * h_src[src_e,h,:]
  *
  * Backward recomputes s_e from the saved (m, Z) statistics:
- *   dh_src[src_e]   += p_e * dout_t                  (fp32 atomics)
- *   ds_e             = p_e * (<dout_t, h_src_e> - <dout_t, out_t>)
- *   dalpha_src/dst  += ds_e * leaky'(s_pre_e)
+ *   dh_src[src_e]  += p_e * dout_t + ds_e * att_src[h]   (fp32 atomics)
+ *   ds_e            = p_e * (<dout_t, h_src_e> - <dout_t, out_t>)
+ *                     * leaky'(s_pre_e)
+ *   datt_src[h]    += ds_e * h_src[src_e];  dh_tgt[t] += (sum ds) * att_dst
+ *   datt_dst[h]    += (sum_e ds_e) * h_tgt[t]
  */
 #include "hip_common.h"
 #include "../include/common.h"
@@ -26,10 +32,17 @@ namespace glt {
 
 namespace {
 
+__device__ __forceinline__ float wave_sum(float v) {
+#pragma unroll
+  for (int sft = kWave / 2; sft > 0; sft >>= 1) v += __shfl_down(v, sft);
+  return __shfl(v, 0);
+}
+
 __global__ void gat_fused_fwd_kernel(
+    const float* __restrict__ h_tgt,      // [Nt.., H, C]
     const float* __restrict__ h_src,      // [Ns, H, C]
-    const float* __restrict__ a_src,      // [Ns, H]
-    const float* __restrict__ a_dst,      // [Nt, H]
+    const float* __restrict__ att_src,    // [H, C]
+    const float* __restrict__ att_dst,    // [H, C]
     const int64_t* __restrict__ src,      // [E]
     const int64_t* __restrict__ offsets,  // [Nt+1]
     int64_t n_tgt, int64_t H, int64_t C, float slope,
@@ -45,12 +58,22 @@ __global__ void gat_fused_fwd_kernel(
     const int64_t t = w / H;
     const int64_t h = w - t * H;
     const int64_t s0 = offsets[t], s1 = offsets[t + 1];
-    const float ad = a_dst[t * H + h];
+    const float* ats = att_src + h * C;
+    const float* atd = att_dst + h * C;
+    const float as0 = lane < C ? ats[lane] : 0.f;
+    const float as1 = kWave + lane < C ? ats[kWave + lane] : 0.f;
+    const float* tv = h_tgt + (t * H + h) * C;
+    const float ad = wave_sum(
+        (lane < C ? tv[lane] * atd[lane] : 0.f) +
+        (kWave + lane < C ? tv[kWave + lane] * atd[kWave + lane] : 0.f));
     float m = -1e30f, Z = 0.f;
     float acc0 = 0.f, acc1 = 0.f;  // lanes cover C (up to 2 passes)
     for (int64_t e = s0; e < s1; ++e) {
       const int64_t sn = src[e];
-      float s = ad + a_src[sn * H + h];
+      const float* hv = h_src + (sn * H + h) * C;
+      const float h0 = lane < C ? hv[lane] : 0.f;
+      const float h1 = kWave + lane < C ? hv[kWave + lane] : 0.f;
+      float s = ad + wave_sum(h0 * as0 + h1 * as1);
       s = s > 0.f ? s : s * slope;
       float scale = 1.f;
       float p;
@@ -62,9 +85,8 @@ __global__ void gat_fused_fwd_kernel(
         p = __expf(s - m);
       }
       Z = Z * scale + p;
-      const float* hv = h_src + (sn * H + h) * C;
-      if (lane < C) acc0 = acc0 * scale + p * hv[lane];
-      if (kWave + lane < C) acc1 = acc1 * scale + p * hv[kWave + lane];
+      acc0 = acc0 * scale + p * h0;
+      acc1 = acc1 * scale + p * h1;
     }
     const float inv = Z > 0.f ? 1.f / Z : 0.f;
     float* ov = out + (t * H + h) * C;
@@ -78,13 +100,14 @@ __global__ void gat_fused_fwd_kernel(
 }
 
 __global__ void gat_fused_bwd_kernel(
-    const float* __restrict__ h_src, const float* __restrict__ a_src,
-    const float* __restrict__ a_dst, const int64_t* __restrict__ src,
-    const int64_t* __restrict__ offsets, const float* __restrict__ out,
-    const float* __restrict__ m_in, const float* __restrict__ z_in,
-    const float* __restrict__ dout, int64_t n_tgt, int64_t H, int64_t C,
-    float slope, float* __restrict__ dh_src, float* __restrict__ da_src,
-    float* __restrict__ da_dst) {
+    const float* __restrict__ h_tgt, const float* __restrict__ h_src,
+    const float* __restrict__ att_src, const float* __restrict__ att_dst,
+    const int64_t* __restrict__ src, const int64_t* __restrict__ offsets,
+    const float* __restrict__ out, const float* __restrict__ m_in,
+    const float* __restrict__ z_in, const float* __restrict__ dout,
+    int64_t n_tgt, int64_t H, int64_t C, float slope,
+    float* __restrict__ dh_tgt, float* __restrict__ dh_src,
+    float* __restrict__ datt_src, float* __restrict__ datt_dst) {
   const int lane = threadIdx.x & (kWave - 1);
   const int64_t wave =
       (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) / kWave;
@@ -95,43 +118,59 @@ __global__ void gat_fused_bwd_kernel(
     const int64_t h = w - t * H;
     const int64_t s0 = offsets[t], s1 = offsets[t + 1];
     if (s1 <= s0) continue;
-    const float ad = a_dst[t * H + h];
+    const float* ats = att_src + h * C;
+    const float* atd = att_dst + h * C;
+    const float as0 = lane < C ? ats[lane] : 0.f;
+    const float as1 = kWave + lane < C ? ats[kWave + lane] : 0.f;
+    const float* tv = h_tgt + (t * H + h) * C;
+    const float t0 = lane < C ? tv[lane] : 0.f;
+    const float t1 = kWave + lane < C ? tv[kWave + lane] : 0.f;
+    const float ad0 = lane < C ? atd[lane] : 0.f;
+    const float ad1 = kWave + lane < C ? atd[kWave + lane] : 0.f;
+    const float ad = wave_sum(t0 * ad0 + t1 * ad1);
     const float m = m_in[t * H + h];
     const float Z = z_in[t * H + h];
     const float inv = Z > 0.f ? 1.f / Z : 0.f;
     const float* dv = dout + (t * H + h) * C;
     const float* ov = out + (t * H + h) * C;
-    float d0 = lane < C ? dv[lane] : 0.f;
-    float d1 = kWave + lane < C ? dv[kWave + lane] : 0.f;
-    // <dout, out>
-    float dot_o = (lane < C ? d0 * ov[lane] : 0.f) +
-                  (kWave + lane < C ? d1 * ov[kWave + lane] : 0.f);
-#pragma unroll
-    for (int sft = kWave / 2; sft > 0; sft >>= 1)
-      dot_o += __shfl_down(dot_o, sft);
-    dot_o = __shfl(dot_o, 0);
-    float dad_acc = 0.f;
+    const float d0 = lane < C ? dv[lane] : 0.f;
+    const float d1 = kWave + lane < C ? dv[kWave + lane] : 0.f;
+    const float dot_o = wave_sum(
+        (lane < C ? d0 * ov[lane] : 0.f) +
+        (kWave + lane < C ? d1 * ov[kWave + lane] : 0.f));
+    float dad_acc = 0.f;        // sum of ds over the segment
+    float das0 = 0.f, das1 = 0.f;  // datt_src accumulator (per lane)
     for (int64_t e = s0; e < s1; ++e) {
       const int64_t sn = src[e];
-      float s_pre = ad + a_src[sn * H + h];
-      float s = s_pre > 0.f ? s_pre : s_pre * slope;
-      const float p = __expf(s - m) * inv;
       const float* hv = h_src + (sn * H + h) * C;
-      float* dhv = dh_src + (sn * H + h) * C;
-      float dot_h = (lane < C ? d0 * hv[lane] : 0.f) +
-                    (kWave + lane < C ? d1 * hv[kWave + lane] : 0.f);
-#pragma unroll
-      for (int sft = kWave / 2; sft > 0; sft >>= 1)
-        dot_h += __shfl_down(dot_h, sft);
-      dot_h = __shfl(dot_h, 0);
-      if (lane < C) atomicAdd(&dhv[lane], p * d0);
-      if (kWave + lane < C) atomicAdd(&dhv[kWave + lane], p * d1);
+      const float h0 = lane < C ? hv[lane] : 0.f;
+      const float h1 = kWave + lane < C ? hv[kWave + lane] : 0.f;
+      const float s_pre = ad + wave_sum(h0 * as0 + h1 * as1);
+      const float s = s_pre > 0.f ? s_pre : s_pre * slope;
+      const float p = __expf(s - m) * inv;
+      const float dot_h = wave_sum(h0 * d0 + h1 * d1);
       float ds = p * (dot_h - dot_o);
       ds *= (s_pre > 0.f ? 1.f : slope);
-      if (lane == 0) atomicAdd(&da_src[sn * H + h], ds);
+      float* dhv = dh_src + (sn * H + h) * C;
+      if (lane < C) atomicAdd(&dhv[lane], p * d0 + ds * as0);
+      if (kWave + lane < C)
+        atomicAdd(&dhv[kWave + lane], p * d1 + ds * as1);
+      das0 += ds * h0;
+      das1 += ds * h1;
       dad_acc += ds;
     }
-    if (lane == 0) atomicAdd(&da_dst[t * H + h], dad_acc);
+    // one atomic flush per wave (not per edge)
+    float* dtv = dh_tgt + (t * H + h) * C;
+    if (lane < C) {
+      atomicAdd(&dtv[lane], dad_acc * ad0);
+      atomicAdd(&datt_src[h * C + lane], das0);
+      atomicAdd(&datt_dst[h * C + lane], dad_acc * t0);
+    }
+    if (kWave + lane < C) {
+      atomicAdd(&dtv[kWave + lane], dad_acc * ad1);
+      atomicAdd(&datt_src[h * C + kWave + lane], das1);
+      atomicAdd(&datt_dst[h * C + kWave + lane], dad_acc * t1);
+    }
   }
 }
 
@@ -144,20 +183,22 @@ int gat_grid(int64_t waves_needed) {
 }  // namespace
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> hip_gat_fused_fwd(
-    const torch::Tensor& h_src, const torch::Tensor& a_src,
-    const torch::Tensor& a_dst, const torch::Tensor& src,
-    const torch::Tensor& offsets, double slope) {
+    const torch::Tensor& h_tgt, const torch::Tensor& h_src,
+    const torch::Tensor& att_src, const torch::Tensor& att_dst,
+    const torch::Tensor& src, const torch::Tensor& offsets, double slope) {
   const int64_t n_tgt = offsets.numel() - 1;
   const int64_t H = h_src.size(1), C = h_src.size(2);
   TORCH_CHECK(C <= 2 * kWave, "GAT fused kernel supports C <= 128");
+  TORCH_CHECK(h_tgt.size(0) >= n_tgt, "h_tgt must cover all targets");
   auto out = torch::empty({n_tgt, H, C}, h_src.options());
   auto m = torch::empty({n_tgt, H}, h_src.options());
   auto z = torch::empty({n_tgt, H}, h_src.options());
   if (n_tgt > 0) {
     hipLaunchKernelGGL(gat_fused_fwd_kernel, dim3(gat_grid(n_tgt * H)),
                        dim3(kBlock), 0, current_stream(),
-                       h_src.data_ptr<float>(), a_src.data_ptr<float>(),
-                       a_dst.data_ptr<float>(), src.data_ptr<int64_t>(),
+                       h_tgt.data_ptr<float>(), h_src.data_ptr<float>(),
+                       att_src.data_ptr<float>(),
+                       att_dst.data_ptr<float>(), src.data_ptr<int64_t>(),
                        offsets.data_ptr<int64_t>(), n_tgt, H, C,
                        (float)slope, out.data_ptr<float>(),
                        m.data_ptr<float>(), z.data_ptr<float>());
@@ -165,30 +206,34 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> hip_gat_fused_fwd(
   return {out, m, z};
 }
 
-std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> hip_gat_fused_bwd(
-    const torch::Tensor& h_src, const torch::Tensor& a_src,
-    const torch::Tensor& a_dst, const torch::Tensor& src,
-    const torch::Tensor& offsets, const torch::Tensor& out,
-    const torch::Tensor& m, const torch::Tensor& z,
-    const torch::Tensor& dout, double slope) {
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
+hip_gat_fused_bwd(const torch::Tensor& h_tgt, const torch::Tensor& h_src,
+                  const torch::Tensor& att_src,
+                  const torch::Tensor& att_dst, const torch::Tensor& src,
+                  const torch::Tensor& offsets, const torch::Tensor& out,
+                  const torch::Tensor& m, const torch::Tensor& z,
+                  const torch::Tensor& dout, double slope) {
   const int64_t n_tgt = offsets.numel() - 1;
   const int64_t H = h_src.size(1), C = h_src.size(2);
-  auto dh = torch::zeros_like(h_src);
-  auto das = torch::zeros_like(a_src);
-  auto dad = torch::zeros_like(a_dst);
+  auto dh_tgt = torch::zeros_like(h_tgt);
+  auto dh_src = torch::zeros_like(h_src);
+  auto das = torch::zeros_like(att_src);
+  auto dad = torch::zeros_like(att_dst);
   if (n_tgt > 0) {
     hipLaunchKernelGGL(gat_fused_bwd_kernel, dim3(gat_grid(n_tgt * H)),
                        dim3(kBlock), 0, current_stream(),
-                       h_src.data_ptr<float>(), a_src.data_ptr<float>(),
-                       a_dst.data_ptr<float>(), src.data_ptr<int64_t>(),
+                       h_tgt.data_ptr<float>(), h_src.data_ptr<float>(),
+                       att_src.data_ptr<float>(),
+                       att_dst.data_ptr<float>(), src.data_ptr<int64_t>(),
                        offsets.data_ptr<int64_t>(),
                        out.data_ptr<float>(), m.data_ptr<float>(),
                        z.data_ptr<float>(),
                        dout.contiguous().data_ptr<float>(), n_tgt, H, C,
-                       (float)slope, dh.data_ptr<float>(),
-                       das.data_ptr<float>(), dad.data_ptr<float>());
+                       (float)slope, dh_tgt.data_ptr<float>(),
+                       dh_src.data_ptr<float>(), das.data_ptr<float>(),
+                       dad.data_ptr<float>());
   }
-  return {dh, das, dad};
+  return {dh_tgt, dh_src, das, dad};
 }
 
 }  // namespace glt

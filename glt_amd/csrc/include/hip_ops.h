@@ -96,16 +96,18 @@ torch::Tensor hip_segment_mean_cat_bwd(const torch::Tensor& dy,
                                        int64_t n_src);
 
 // --- fused GAT edge softmax + aggregation (hip_gat.hip) ---------------------
+// Attention logits are computed inside the kernels from (h, att) directly.
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> hip_gat_fused_fwd(
-    const torch::Tensor& h_src, const torch::Tensor& a_src,
-    const torch::Tensor& a_dst, const torch::Tensor& src,
-    const torch::Tensor& offsets, double slope);
-std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> hip_gat_fused_bwd(
-    const torch::Tensor& h_src, const torch::Tensor& a_src,
-    const torch::Tensor& a_dst, const torch::Tensor& src,
-    const torch::Tensor& offsets, const torch::Tensor& out,
-    const torch::Tensor& m, const torch::Tensor& z,
-    const torch::Tensor& dout, double slope);
+    const torch::Tensor& h_tgt, const torch::Tensor& h_src,
+    const torch::Tensor& att_src, const torch::Tensor& att_dst,
+    const torch::Tensor& src, const torch::Tensor& offsets, double slope);
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
+hip_gat_fused_bwd(const torch::Tensor& h_tgt, const torch::Tensor& h_src,
+                  const torch::Tensor& att_src,
+                  const torch::Tensor& att_dst, const torch::Tensor& src,
+                  const torch::Tensor& offsets, const torch::Tensor& out,
+                  const torch::Tensor& m, const torch::Tensor& z,
+                  const torch::Tensor& dout, double slope);
 
 // --- f32 MFMA projection GEMM (hip_gemm_f32.hip) ----------------------------
 torch::Tensor hip_sage_gemm(const torch::Tensor& A, const torch::Tensor& B,

@@ -157,23 +157,27 @@ class GATConv(nn.Module):
         relations sharing a node type into one GEMM)."""
         tgt, src = edge_index[0], edge_index[1]
         h = h_src
-        alpha_src = (h_src * self.att_src).sum(-1)
-        alpha_dst = (h_tgt * self.att_dst).sum(-1)
         if (getattr(self, "use_fused", True) and h_src.is_cuda
                 and h_src.dtype == torch.float32
                 and self.out_channels <= 128
                 and not (self.training and self.dropout > 0)):
-            # fused segment-softmax-aggregate (edges sorted by target)
+            # fused segment-softmax-aggregate (edges sorted by target);
+            # the kernel computes the attention logits itself, so no
+            # per-node alpha tensors are built here
             from ..ops import gat_softmax_aggregate
 
-            out = gat_softmax_aggregate(h_src, alpha_src,
-                                        alpha_dst[:nt], tgt, src, nt,
+            out = gat_softmax_aggregate(h_tgt, h_src,
+                                        self.att_src.squeeze(0),
+                                        self.att_dst.squeeze(0),
+                                        tgt, src, nt,
                                         self.negative_slope)
             out = out.reshape(nt, self.heads * self.out_channels) \
                 if self.concat else out.mean(dim=1)
             if self.bias is not None:
                 out = out + self.bias
             return out
+        alpha_src = (h_src * self.att_src).sum(-1)
+        alpha_dst = (h_tgt * self.att_dst).sum(-1)
         # index_select (not advanced indexing): its backward is an
         # index_add scatter, avoiding the radix-sort the indexing backward
         # performs per gather (24 device sorts/step in RGAT otherwise)

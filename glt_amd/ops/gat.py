@@ -1,19 +1,26 @@
-"""Autograd wrapper for the fused GAT edge-softmax + aggregation kernels."""
+"""Autograd wrapper for the fused GAT edge-softmax + aggregation kernels.
+
+The kernels compute the attention logits internally from (h, att), so the
+python layer never materializes per-node alpha tensors (saves ~8 small
+launches per relation per layer in RGAT's launch-bound regime).
+"""
 import torch
 
 
 class _GatFused(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, h_src, a_src, a_dst, src, offsets, slope):
+    def forward(ctx, h_tgt, h_src, att_src, att_dst, src, offsets, slope):
         from .. import _C
 
         # contiguity once, shared by fwd and the saved tensors bwd reads
+        h_tgt = h_tgt.contiguous()
         h_src = h_src.contiguous()
-        a_src = a_src.contiguous()
-        a_dst = a_dst.contiguous()
-        out, m, z = _C.gat_fused_fwd(h_src, a_src, a_dst, src, offsets,
-                                     slope)
-        ctx.save_for_backward(h_src, a_src, a_dst, src, offsets, out, m, z)
+        att_src = att_src.contiguous()
+        att_dst = att_dst.contiguous()
+        out, m, z = _C.gat_fused_fwd(h_tgt, h_src, att_src, att_dst, src,
+                                     offsets, slope)
+        ctx.save_for_backward(h_tgt, h_src, att_src, att_dst, src, offsets,
+                              out, m, z)
         ctx.slope = slope
         return out
 
@@ -21,18 +28,21 @@ class _GatFused(torch.autograd.Function):
     def backward(ctx, dout):
         from .. import _C
 
-        h_src, a_src, a_dst, src, offsets, out, m, z = ctx.saved_tensors
-        dh, das, dad = _C.gat_fused_bwd(h_src, a_src, a_dst, src, offsets,
-                                        out, m, z, dout, ctx.slope)
-        return dh, das, dad, None, None, None
+        (h_tgt, h_src, att_src, att_dst, src, offsets, out, m,
+         z) = ctx.saved_tensors
+        dht, dhs, das, dad = _C.gat_fused_bwd(h_tgt, h_src, att_src,
+                                              att_dst, src, offsets, out,
+                                              m, z, dout, ctx.slope)
+        return dht, dhs, das, dad, None, None, None
 
 
-def gat_softmax_aggregate(h_src, a_src, a_dst, tgt, src, n_tgt,
+def gat_softmax_aggregate(h_tgt, h_src, att_src, att_dst, tgt, src, n_tgt,
                           negative_slope=0.2):
-    """out[t,h,:] = sum_e softmax_t(leaky(a_dst[t,h]+a_src[src_e,h]))
-    * h_src[src_e,h,:], over edges sorted by target."""
+    """out[t,h,:] = sum_e softmax_t(leaky(<h_tgt[t,h],att_dst[h]> +
+    <h_src[src_e,h],att_src[h]>)) * h_src[src_e,h,:], edges sorted by
+    target.  att_* are [H, C]."""
     from .segment import _boundaries
 
     offsets = torch.searchsorted(tgt, _boundaries(n_tgt, tgt.device))
-    return _GatFused.apply(h_src, a_src, a_dst, src.contiguous(), offsets,
-                           negative_slope)
+    return _GatFused.apply(h_tgt, h_src, att_src, att_dst,
+                           src.contiguous(), offsets, negative_slope)
