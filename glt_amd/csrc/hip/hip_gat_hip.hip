@@ -48,7 +48,8 @@ __global__ void gat_fused_fwd_kernel(
     int64_t n_tgt, int64_t H, int64_t C, float slope,
     float* __restrict__ out,              // [Nt, H, C]
     float* __restrict__ m_out,            // [Nt, H]
-    float* __restrict__ z_out) {          // [Nt, H]
+    float* __restrict__ z_out,            // [Nt, H]
+    float* __restrict__ spre_out) {       // [E, H] pre-activation logits
   const int lane = threadIdx.x & (kWave - 1);
   const int64_t wave =
       (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) / kWave;
@@ -74,6 +75,7 @@ __global__ void gat_fused_fwd_kernel(
       const float h0 = lane < C ? hv[lane] : 0.f;
       const float h1 = kWave + lane < C ? hv[kWave + lane] : 0.f;
       float s = ad + wave_sum(h0 * as0 + h1 * as1);
+      if (lane == 0) spre_out[e * H + h] = s;  // backward reuses this
       s = s > 0.f ? s : s * slope;
       float scale = 1.f;
       float p;
@@ -104,7 +106,8 @@ __global__ void gat_fused_bwd_kernel(
     const float* __restrict__ att_src, const float* __restrict__ att_dst,
     const int64_t* __restrict__ src, const int64_t* __restrict__ offsets,
     const float* __restrict__ out, const float* __restrict__ m_in,
-    const float* __restrict__ z_in, const float* __restrict__ dout,
+    const float* __restrict__ z_in, const float* __restrict__ spre_in,
+    const float* __restrict__ dout,
     int64_t n_tgt, int64_t H, int64_t C, float slope,
     float* __restrict__ dh_tgt, float* __restrict__ dh_src,
     float* __restrict__ datt_src, float* __restrict__ datt_dst) {
@@ -127,7 +130,6 @@ __global__ void gat_fused_bwd_kernel(
     const float t1 = kWave + lane < C ? tv[kWave + lane] : 0.f;
     const float ad0 = lane < C ? atd[lane] : 0.f;
     const float ad1 = kWave + lane < C ? atd[kWave + lane] : 0.f;
-    const float ad = wave_sum(t0 * ad0 + t1 * ad1);
     const float m = m_in[t * H + h];
     const float Z = z_in[t * H + h];
     const float inv = Z > 0.f ? 1.f / Z : 0.f;
@@ -145,7 +147,8 @@ __global__ void gat_fused_bwd_kernel(
       const float* hv = h_src + (sn * H + h) * C;
       const float h0 = lane < C ? hv[lane] : 0.f;
       const float h1 = kWave + lane < C ? hv[kWave + lane] : 0.f;
-      const float s_pre = ad + wave_sum(h0 * as0 + h1 * as1);
+      // logits were cached by the forward pass: no recompute chain here
+      const float s_pre = spre_in[e * H + h];
       const float s = s_pre > 0.f ? s_pre : s_pre * slope;
       const float p = __expf(s - m) * inv;
       const float dot_h = wave_sum(h0 * d0 + h1 * d1);
@@ -182,17 +185,20 @@ int gat_grid(int64_t waves_needed) {
 
 }  // namespace
 
-std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> hip_gat_fused_fwd(
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
+hip_gat_fused_fwd(
     const torch::Tensor& h_tgt, const torch::Tensor& h_src,
     const torch::Tensor& att_src, const torch::Tensor& att_dst,
     const torch::Tensor& src, const torch::Tensor& offsets, double slope) {
   const int64_t n_tgt = offsets.numel() - 1;
   const int64_t H = h_src.size(1), C = h_src.size(2);
+  const int64_t E = src.numel();
   TORCH_CHECK(C <= 2 * kWave, "GAT fused kernel supports C <= 128");
   TORCH_CHECK(h_tgt.size(0) >= n_tgt, "h_tgt must cover all targets");
   auto out = torch::empty({n_tgt, H, C}, h_src.options());
   auto m = torch::empty({n_tgt, H}, h_src.options());
   auto z = torch::empty({n_tgt, H}, h_src.options());
+  auto spre = torch::empty({E, H}, h_src.options());
   if (n_tgt > 0) {
     hipLaunchKernelGGL(gat_fused_fwd_kernel, dim3(gat_grid(n_tgt * H)),
                        dim3(kBlock), 0, current_stream(),
@@ -201,9 +207,10 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> hip_gat_fused_fwd(
                        att_dst.data_ptr<float>(), src.data_ptr<int64_t>(),
                        offsets.data_ptr<int64_t>(), n_tgt, H, C,
                        (float)slope, out.data_ptr<float>(),
-                       m.data_ptr<float>(), z.data_ptr<float>());
+                       m.data_ptr<float>(), z.data_ptr<float>(),
+                       spre.data_ptr<float>());
   }
-  return {out, m, z};
+  return {out, m, z, spre};
 }
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
@@ -212,7 +219,8 @@ hip_gat_fused_bwd(const torch::Tensor& h_tgt, const torch::Tensor& h_src,
                   const torch::Tensor& att_dst, const torch::Tensor& src,
                   const torch::Tensor& offsets, const torch::Tensor& out,
                   const torch::Tensor& m, const torch::Tensor& z,
-                  const torch::Tensor& dout, double slope) {
+                  const torch::Tensor& spre, const torch::Tensor& dout,
+                  double slope) {
   const int64_t n_tgt = offsets.numel() - 1;
   const int64_t H = h_src.size(1), C = h_src.size(2);
   auto dh_tgt = torch::zeros_like(h_tgt);
@@ -227,7 +235,7 @@ hip_gat_fused_bwd(const torch::Tensor& h_tgt, const torch::Tensor& h_src,
                        att_dst.data_ptr<float>(), src.data_ptr<int64_t>(),
                        offsets.data_ptr<int64_t>(),
                        out.data_ptr<float>(), m.data_ptr<float>(),
-                       z.data_ptr<float>(),
+                       z.data_ptr<float>(), spre.data_ptr<float>(),
                        dout.contiguous().data_ptr<float>(), n_tgt, H, C,
                        (float)slope, dh_tgt.data_ptr<float>(),
                        dh_src.data_ptr<float>(), das.data_ptr<float>(),

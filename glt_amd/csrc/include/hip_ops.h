@@ -97,7 +97,9 @@ torch::Tensor hip_segment_mean_cat_bwd(const torch::Tensor& dy,
 
 // --- fused GAT edge softmax + aggregation (hip_gat.hip) ---------------------
 // Attention logits are computed inside the kernels from (h, att) directly.
-std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> hip_gat_fused_fwd(
+// fwd returns (out, m, Z, s_pre); bwd consumes the cached s_pre logits.
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
+hip_gat_fused_fwd(
     const torch::Tensor& h_tgt, const torch::Tensor& h_src,
     const torch::Tensor& att_src, const torch::Tensor& att_dst,
     const torch::Tensor& src, const torch::Tensor& offsets, double slope);
@@ -107,7 +109,8 @@ hip_gat_fused_bwd(const torch::Tensor& h_tgt, const torch::Tensor& h_src,
                   const torch::Tensor& att_dst, const torch::Tensor& src,
                   const torch::Tensor& offsets, const torch::Tensor& out,
                   const torch::Tensor& m, const torch::Tensor& z,
-                  const torch::Tensor& dout, double slope);
+                  const torch::Tensor& spre, const torch::Tensor& dout,
+                  double slope);
 
 // --- f32 MFMA projection GEMM (hip_gemm_f32.hip) ----------------------------
 torch::Tensor hip_sage_gemm(const torch::Tensor& A, const torch::Tensor& B,

@@ -17,10 +17,10 @@ class _GatFused(torch.autograd.Function):
         h_src = h_src.contiguous()
         att_src = att_src.contiguous()
         att_dst = att_dst.contiguous()
-        out, m, z = _C.gat_fused_fwd(h_tgt, h_src, att_src, att_dst, src,
-                                     offsets, slope)
+        out, m, z, spre = _C.gat_fused_fwd(h_tgt, h_src, att_src, att_dst,
+                                           src, offsets, slope)
         ctx.save_for_backward(h_tgt, h_src, att_src, att_dst, src, offsets,
-                              out, m, z)
+                              out, m, z, spre)
         ctx.slope = slope
         return out
 
@@ -28,11 +28,11 @@ class _GatFused(torch.autograd.Function):
     def backward(ctx, dout):
         from .. import _C
 
-        (h_tgt, h_src, att_src, att_dst, src, offsets, out, m,
-         z) = ctx.saved_tensors
+        (h_tgt, h_src, att_src, att_dst, src, offsets, out, m, z,
+         spre) = ctx.saved_tensors
         dht, dhs, das, dad = _C.gat_fused_bwd(h_tgt, h_src, att_src,
                                               att_dst, src, offsets, out,
-                                              m, z, dout, ctx.slope)
+                                              m, z, spre, dout, ctx.slope)
         return dht, dhs, das, dad, None, None, None
 
 
