@@ -1,0 +1,106 @@
+"""Focused unit tests for smaller components (partition books, dist
+context role math, typing helpers, sampler dataclasses, pygcompat)."""
+import torch
+
+import glt_amd
+
+
+def test_range_partition_book():
+    from glt_amd.partition import RangePartitionBook
+
+    # partitions: [0,10), [10,25), [25,40)
+    bounds = torch.tensor([10, 25, 40])
+    pb1 = RangePartitionBook(bounds, partition_idx=1)
+    ids = torch.tensor([0, 9, 10, 24, 25, 39])
+    assert pb1[ids].tolist() == [0, 0, 1, 1, 2, 2]
+    assert len(pb1) == 40
+    assert pb1.offset == 10
+    assert pb1.id2index(torch.tensor([10, 24])).tolist() == [0, 14]
+    pb0 = RangePartitionBook(bounds, partition_idx=0)
+    assert pb0.offset == 0
+
+
+def test_glt_partition_book():
+    from glt_amd.partition import GLTPartitionBook
+
+    book = GLTPartitionBook(torch.tensor([0, 1, 1, 2, 0]))
+    out = book[torch.tensor([1, 3, 4])]
+    assert out.tolist() == [1, 2, 0]
+    assert out.dtype == torch.long
+    assert len(book) == 5
+
+
+def test_dist_context_roles():
+    from glt_amd.distributed.dist_context import (
+        DistContext, DistRole, _set_client_context, _set_server_context,
+        assign_server_by_order, get_context, init_worker_group)
+
+    init_worker_group(world_size=4, rank=2)
+    ctx = get_context()
+    assert ctx.is_worker and not ctx.is_server
+    assert ctx.global_rank == 2 and ctx.worker_name.endswith("_2")
+
+    _set_server_context(num_servers=2, server_rank=1, num_clients=3)
+    ctx = get_context()
+    assert ctx.is_server and ctx.global_world_size == 5
+    assert ctx.global_rank == 1
+
+    _set_client_context(num_servers=2, num_clients=3, client_rank=0)
+    ctx = get_context()
+    assert ctx.is_client and ctx.global_rank == 2  # after the servers
+
+    # more clients than servers: round robin
+    assert assign_server_by_order(4, num_servers=3, num_clients=6) == [1]
+    # fewer clients than servers: contiguous spans covering all servers
+    spans = [assign_server_by_order(c, 5, 2) for c in range(2)]
+    assert sorted(s for span in spans for s in span) == [0, 1, 2, 3, 4]
+
+
+def test_negative_sampling_semantics():
+    from glt_amd.sampler import NegativeSampling
+
+    b = NegativeSampling("binary", amount=0.5)
+    assert b.is_binary() and not b.is_triplet()
+    assert b.amount == 0.5  # binary keeps fractional ratio
+    t = NegativeSampling("triplet", amount=1.5)
+    assert t.is_triplet()
+    assert t.amount == 2  # triplet rounds up to whole negatives per pos
+
+
+def test_reverse_edge_type():
+    from glt_amd.typing import reverse_edge_type
+
+    assert reverse_edge_type(("u", "r", "v")) == ("v", "rev_r", "u")
+    # double reverse returns the original
+    assert reverse_edge_type(reverse_edge_type(("u", "r", "v"))) == \
+        ("u", "r", "v")
+
+
+def test_topology_layouts(ring_graph):
+    from glt_amd.data import Topology
+
+    topo = Topology(ring_graph["edge_index"], num_nodes=40)
+    assert topo.indptr.numel() == 41
+    rows, cols, eids = topo.to_coo()
+    got = set(zip(rows.tolist(), cols.tolist()))
+    want = set(zip(ring_graph["edge_index"][0].tolist(),
+                   ring_graph["edge_index"][1].tolist()))
+    assert got == want
+    assert torch.equal(topo.degrees[torch.tensor([0, 7])],
+                       torch.tensor([2, 2]))
+
+
+def test_apply_to_tensors():
+    from glt_amd.pygcompat.data import Data, HeteroData, apply_to_tensors
+
+    d = Data(x=torch.ones(3, 2), edge_index=torch.zeros(2, 4,
+                                                        dtype=torch.long))
+    d2 = apply_to_tensors(d, lambda t: t * 2)
+    assert torch.equal(d2.x, torch.full((3, 2), 2.0))
+
+    h = HeteroData()
+    h["u"].x = torch.ones(2, 2)
+    h[("u", "e", "v")].edge_index = torch.zeros(2, 3, dtype=torch.long)
+    h2 = apply_to_tensors(h, lambda t: t + 1)
+    assert torch.equal(h2["u"].x, torch.full((2, 2), 2.0))
+    assert h2[("u", "e", "v")].edge_index.max().item() == 1
