@@ -6,12 +6,11 @@ and benchmarks need, written against the glt_amd batch convention:
 ``edge_index[0]`` = target-side (seed) local index, ``edge_index[1]`` =
 source/neighbor local index; aggregation flows 1 -> 0.
 
-All dense math lands on rocBLAS/hipBLASLt through torch.nn.Linear; the
-sparse aggregation uses index_add_ (rocPRIM-backed scatter) — a fused
-CSR-segment kernel is a planned optimization.
+Dense math lands on hipBLASLt through torch.nn.Linear (or the in-house
+MFMA GEMM for skinny shapes); sparse aggregation uses the fused HIP
+kernels in csrc/hip/ (segment-mean [+root concat], GAT edge-softmax) on
+GPU fp32 batches, with an index_add_ fallback for CPU/other dtypes.
 """
-import math
-from typing import Optional, Tuple
 
 import torch
 import torch.nn as nn

@@ -1,6 +1,6 @@
 """Model families exercised by the reference examples: GraphSAGE
 (supervised + unsupervised link-pred), GAT, GCN."""
-from typing import List, Optional
+from typing import Optional
 
 import torch
 import torch.nn as nn

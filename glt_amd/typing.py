@@ -1,5 +1,5 @@
 """Common type aliases (parity: reference python/typing.py)."""
-from typing import Dict, List, Optional, Tuple, Union
+from typing import Dict, Tuple, Union
 
 import torch
 
