@@ -18,7 +18,9 @@ ROOT = os.path.dirname(os.path.abspath(__file__))
 sources = sorted(
     glob.glob("glt_amd/csrc/*.cpp")
     + glob.glob("glt_amd/csrc/cpu/*.cpp")
-    + glob.glob("glt_amd/csrc/hip/*.hip")
+    + [f for f in glob.glob("glt_amd/csrc/hip/*.hip")
+       # exclude the *_hip.hip copies torch's hipify step regenerates
+       if not f.endswith("_hip.hip")]
 )
 
 setup(
