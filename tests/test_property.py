@@ -114,3 +114,53 @@ def test_stitch_matches_python_reference(n_ids, parts, rnd):
     flat_ref = [v for vals in expect for v in (vals or [])]
     assert nbrs.tolist() == flat_ref
     assert num.tolist() == [len(v or []) for v in expect]
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.integers(2, 50), st.integers(0, 200), st.integers(-1, 8),
+       st.randoms())
+def test_cpu_sampler_invariants(n, e, k, rnd):
+    """For ANY random CSR + seeds + fan-out: counts = min(k, deg) (or deg
+    when k=-1), every sampled neighbor is a true neighbor of its row, and
+    uniform sampling is without replacement."""
+    if k == 0:
+        k = 1
+    rows = torch.tensor([rnd.randrange(n) for _ in range(e)],
+                        dtype=torch.long)
+    cols = torch.tensor([rnd.randrange(n) for _ in range(e)],
+                        dtype=torch.long)
+    topo = Topology(torch.stack([rows, cols]) if e else
+                    torch.empty(2, 0, dtype=torch.long), num_nodes=n)
+    seeds = torch.tensor([rnd.randrange(n) for _ in range(5)],
+                         dtype=torch.long)
+    nbrs, num, _ = _C.sample_neighbors(topo.indptr, topo.indices, seeds, k)
+    adj = {}
+    for r, c in zip(rows.tolist(), cols.tolist()):
+        adj.setdefault(r, []).append(c)
+    off = 0
+    for s, cnt in zip(seeds.tolist(), num.tolist()):
+        deg = len(adj.get(s, []))
+        expect = deg if k == -1 else min(k, deg)
+        assert cnt == expect, (s, cnt, expect)
+        picked = nbrs[off:off + cnt].tolist()
+        off += cnt
+        # without replacement over EDGES: each value appears at most as
+        # often as its multiplicity in the (multi)graph's adjacency
+        import collections
+        have = collections.Counter(adj.get(s, []))
+        for v, c in collections.Counter(picked).items():
+            assert c <= have[v], (s, v, c, have[v])
+    assert off == nbrs.numel()
+
+
+@settings(max_examples=20, deadline=None)
+@given(st.integers(3, 40), st.integers(1, 120), st.randoms())
+def test_negative_sampler_strict_property(n, e, rnd):
+    """Strict negatives never collide with an existing edge."""
+    rows = torch.tensor([rnd.randrange(n) for _ in range(e)])
+    cols = torch.tensor([rnd.randrange(n) for _ in range(e)])
+    topo = Topology(torch.stack([rows, cols]), num_nodes=n)
+    neg = _C.sample_negative(topo.indptr, topo.indices, n, 32, 8, False)
+    present = set(zip(rows.tolist(), cols.tolist()))
+    for r, c in neg.t().tolist():
+        assert (r, c) not in present
