@@ -242,7 +242,6 @@ class DeferredSampler {
       cols_.push_back(torch::empty({ec}, opts));
       uniq_.push_back(torch::empty({ec}, opts));
       flags_.push_back(torch::empty({ec}, opts));
-      frontier_n_.push_back(torch::zeros({1}, opts));
     }
     count_dev_ = torch::zeros({1}, opts);
     seed_uniq_ = torch::empty({batch_cap}, opts);
@@ -266,6 +265,8 @@ class DeferredSampler {
                 "deferred: seeds must be int64");
     TORCH_CHECK(seeds_in.device() == device_,
                 "deferred: seeds on wrong device");
+    TORCH_CHECK(!with_eid_ || edge_ids.has_value(),
+                "deferred: with_eid sampler needs edge_ids");
     auto seeds = seeds_in.contiguous();
     const int64_t num_rows = indptr.size(0) - 1;
     const int64_t bs = seeds.size(0);
@@ -405,7 +406,7 @@ class DeferredSampler {
   torch::Tensor keys_, first_idx_, local_id_, count_dev_;
   torch::Tensor seed_uniq_, seed_flags_, seed_n_, stats_;
   std::vector<torch::Tensor> counts_, offsets_, nbrs_, eids_, rows_, cols_,
-      uniq_, flags_, frontier_n_;
+      uniq_, flags_;
 };
 
 std::shared_ptr<DeferredSampler> deferred_sampler_create(
