@@ -161,3 +161,51 @@ def test_shm_channel_multithread_stress():
     for t in range(n_threads):
         seq = [i for tid, i in got if tid == t]
         assert seq == sorted(seq)
+
+
+def test_coo_to_csr_edge_cases():
+    from glt_amd.utils.topo import coo_to_csr
+
+    # empty graph
+    indptr, indices, eids, ew = coo_to_csr(
+        torch.empty(0, dtype=torch.long), torch.empty(0, dtype=torch.long),
+        num_rows=5)
+    assert indptr.tolist() == [0] * 6 and indices.numel() == 0
+    # isolated rows + per-row sorted payloads stay aligned
+    row = torch.tensor([3, 0, 3, 0])
+    col = torch.tensor([7, 1, 5, 2])
+    eid = torch.tensor([10, 11, 12, 13])
+    ew = torch.tensor([0.1, 0.2, 0.3, 0.4])
+    indptr, indices, eids, ews = coo_to_csr(row, col, eid, ew, num_rows=4)
+    assert indptr.tolist() == [0, 2, 2, 2, 4]
+    # edge payloads must follow their (row, col) through the sort
+    want = {(0, 1): (11, 0.2), (0, 2): (13, 0.4),
+            (3, 7): (10, 0.1), (3, 5): (12, 0.3)}
+    for r in range(4):
+        for k in range(indptr[r], indptr[r + 1]):
+            c = indices[k].item()
+            assert (eids[k].item(),
+                    round(ews[k].item(), 6)) == want[(r, c)]
+
+
+def test_message_roundtrip_edge_cases():
+    from glt_amd.distributed.message import (END_KEY,
+                                             decode_sample_message,
+                                             encode_sampler_output)
+    from glt_amd.sampler import SamplerOutput
+
+    # empty edge set (isolated seeds), metadata with tensor + scalar
+    out = SamplerOutput(
+        node=torch.tensor([4, 9]), row=torch.empty(0, dtype=torch.long),
+        col=torch.empty(0, dtype=torch.long), edge=None,
+        batch=torch.tensor([4, 9]), num_sampled_nodes=[2, 0],
+        num_sampled_edges=[0],
+        metadata={"edge_label": torch.tensor([1.0]), "input_type": None})
+    msg = encode_sampler_output(out)
+    assert END_KEY not in msg
+    dec, x, y, ea = decode_sample_message(msg)
+    assert x is None and y is None and ea is None
+    assert torch.equal(dec.node, out.node)
+    assert dec.row.numel() == 0
+    assert dec.num_sampled_nodes == [2, 0]
+    assert torch.equal(dec.metadata["edge_label"], torch.tensor([1.0]))
