@@ -10,7 +10,8 @@ from .dist_neighbor_sampler import DistNeighborSampler
 from .dist_options import (CollocatedDistSamplingWorkerOptions,
                            MpDistSamplingWorkerOptions,
                            RemoteDistSamplingWorkerOptions)
-from .dist_random_partitioner import DistRandomPartitioner
+from .dist_random_partitioner import (DistHeteroRandomPartitioner,
+                                      DistRandomPartitioner)
 from .dist_sampling_producer import (DistCollocatedSamplingProducer,
                                      DistMpSamplingProducer)
 from .dist_server import (DistServer, get_server, init_server,
@@ -28,6 +29,7 @@ __all__ = [
     "init_worker_group", "DistDataset", "DistFeature", "DistGraph",
     "DistLoader", "DistLinkNeighborLoader", "DistNeighborLoader",
     "DistSubGraphLoader", "DistNeighborSampler", "DistRandomPartitioner",
+    "DistHeteroRandomPartitioner",
     "CollocatedDistSamplingWorkerOptions", "MpDistSamplingWorkerOptions",
     "RemoteDistSamplingWorkerOptions", "DistCollocatedSamplingProducer",
     "DistMpSamplingProducer", "DistServer", "get_server", "init_server",

@@ -153,3 +153,77 @@ class DistRandomPartitioner:
             save_feature_partition(output_dir, self.rank, feat)
         barrier()
         return pb, graph, feat
+
+
+class DistHeteroRandomPartitioner:
+    """Hetero online partitioning (capability parity: reference
+    dist_random_partitioner.py:300-539): one deterministic node book per
+    node TYPE (so every edge type sharing a src type agrees on ownership),
+    per-edge-type edge exchange and per-node-type feature exchange, all
+    composed from the homo partitioner's RPC push machinery.
+
+    Every rank must construct with the same type sets (callee
+    registration and barriers run in sorted-type order).
+    """
+
+    def __init__(self, num_nodes: Dict[str, int],
+                 local_edge_index: Dict[tuple, torch.Tensor],
+                 local_eids: Optional[Dict[tuple, torch.Tensor]] = None,
+                 local_node_feat: Optional[Dict[str, torch.Tensor]] = None,
+                 local_node_ids: Optional[Dict[str, torch.Tensor]] = None,
+                 edge_assign_strategy: str = "by_src", seed: int = 0):
+        import zlib
+
+        self.edge_types = sorted(local_edge_index.keys())
+        self.node_types = sorted(num_nodes.keys())
+        self.edge_assign_strategy = edge_assign_strategy
+
+        def type_seed(nt: str) -> int:
+            return seed * 0x9E3779B9 + zlib.crc32(nt.encode())
+
+        # per-edge-type partitioners exchange edges; the book each uses is
+        # the ASSIGN-side node type's book (same seed per type across all
+        # edge types -> consistent ownership)
+        self._edge_parts = {}
+        for et in self.edge_types:
+            assign_t = et[0] if edge_assign_strategy == "by_src" else et[2]
+            self._edge_parts[et] = DistRandomPartitioner(
+                num_nodes[assign_t], local_edge_index[et],
+                local_eids.get(et) if local_eids else None,
+                edge_assign_strategy=edge_assign_strategy,
+                seed=type_seed(assign_t))
+        # per-node-type partitioners exchange feature rows
+        self._feat_parts = {}
+        for nt in self.node_types:
+            feat = local_node_feat.get(nt) if local_node_feat else None
+            if feat is None:
+                continue
+            self._feat_parts[nt] = DistRandomPartitioner(
+                num_nodes[nt],
+                torch.empty(2, 0, dtype=torch.long),
+                local_node_feat=feat,
+                local_node_ids=(local_node_ids.get(nt)
+                                if local_node_ids else None),
+                seed=type_seed(nt))
+        self._num_nodes = num_nodes
+        self._seed_of = type_seed
+
+    def node_pbs(self) -> Dict[str, GLTPartitionBook]:
+        out = {}
+        for nt in self.node_types:
+            helper = DistRandomPartitioner.__new__(DistRandomPartitioner)
+            helper.seed = self._seed_of(nt)
+            helper.num_nodes = self._num_nodes[nt]
+            helper.num_parts = get_context().world_size
+            out[nt] = DistRandomPartitioner.node_pb(helper)
+        return out
+
+    def partition(self):
+        """Returns (node_pbs, graph_parts, feat_parts) — all dicts keyed
+        by node/edge type; this rank's shard of each."""
+        graphs, feats = {}, {}
+        for et in self.edge_types:  # same order on every rank
+            _, graphs[et], _ = self._edge_parts[et].partition()
+        for nt in sorted(self._feat_parts.keys()):
+            _, _, feats[nt] = self._feat_parts[nt].partition()
+        return self.node_pbs(), graphs, feats

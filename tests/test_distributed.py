@@ -627,3 +627,69 @@ def test_dist_link_neighbor_loader():
             p.terminate()
     for rank, err in results:
         assert err is None, f"rank {rank}:\n{err}"
+
+
+def _worker_hetero_dist_partitioner(rank, world, port, q):
+    try:
+        import torch
+
+        import glt_amd
+        from glt_amd.distributed import init_worker_group
+        from glt_amd.distributed.rpc import init_rpc, shutdown_rpc
+        from glt_amd.distributed.dist_random_partitioner import \
+            DistHeteroRandomPartitioner
+
+        init_worker_group(world, rank)
+        init_rpc("127.0.0.1", port)
+        n_u, n_v = 60, 40
+        # rank r holds the edges whose src falls in its half
+        u_src = torch.arange(rank * 30, rank * 30 + 30).repeat_interleave(2)
+        v_dst = torch.randint(0, n_v, (60,))
+        v_src = torch.arange(rank * 20, rank * 20 + 20)
+        u_dst = torch.randint(0, n_u, (20,))
+        ei = {("u", "r", "v"): torch.stack([u_src, v_dst]),
+              ("v", "s", "u"): torch.stack([v_src, u_dst])}
+        feats = {"u": torch.arange(rank * 30, rank * 30 + 30,
+                                   dtype=torch.float32).unsqueeze(1)}
+        ids = {"u": torch.arange(rank * 30, rank * 30 + 30)}
+        p = DistHeteroRandomPartitioner(
+            {"u": n_u, "v": n_v}, ei, local_node_feat=feats,
+            local_node_ids=ids, seed=3)
+        pbs, graphs, fp = p.partition()
+        # every received edge is owned here by its src-type book
+        assert (pbs["u"][graphs[("u", "r", "v")].edge_index[0]]
+                == rank).all()
+        assert (pbs["v"][graphs[("v", "s", "u")].edge_index[0]]
+                == rank).all()
+        # feature rows owned here + closed-form values
+        assert (pbs["u"][fp["u"].ids] == rank).all()
+        assert (fp["u"].feats == fp["u"].ids.float().unsqueeze(1)).all()
+        from glt_amd.distributed import barrier
+
+        barrier()
+        shutdown_rpc()
+        q.put((rank, None))
+    except Exception:  # noqa: BLE001
+        import traceback
+
+        q.put((rank, traceback.format_exc()))
+
+
+@pytest.mark.timeout(180)
+def test_dist_hetero_random_partitioner():
+    from glt_amd.utils import get_free_port
+
+    ctx = mp.get_context("spawn")
+    port = get_free_port()
+    q = ctx.Queue()
+    ps = [ctx.Process(target=_worker_hetero_dist_partitioner,
+                      args=(r, 2, port, q)) for r in range(2)]
+    for p in ps:
+        p.start()
+    results = [q.get(timeout=150) for _ in range(2)]
+    for p in ps:
+        p.join(timeout=30)
+        if p.is_alive():
+            p.terminate()
+    for rank, err in results:
+        assert err is None, f"rank {rank}:\n{err}"
