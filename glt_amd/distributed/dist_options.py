@@ -84,7 +84,13 @@ class RemoteDistSamplingWorkerOptions(_BasicDistSamplingWorkerOptions):
         self.buffer_size = buffer_size
         self.buffer_capacity = buffer_capacity
         self.prefetch_size = prefetch_size
-        self.worker_key = worker_key or "default"
+        # accepted for reference signature compatibility; workload_type
+        # namespaces the server-side producer key alongside worker_key
+        self.glt_graph = glt_graph
+        self.workload_type = workload_type
+        self.worker_key = (f"{workload_type}:{worker_key}"
+                           if workload_type and worker_key else
+                           worker_key or workload_type or "default")
 
 
 AllDistSamplingWorkerOptions = Union[CollocatedDistSamplingWorkerOptions,
