@@ -12,3 +12,7 @@ test-gpu:
 
 bench:
 	python bench.py --gpus 1 --steps 60 --warmup 10
+
+lint:
+	python -m compileall -q glt_amd tools bench.py __graft_entry__.py
+	python tools/check_imports.py
