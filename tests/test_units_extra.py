@@ -250,3 +250,20 @@ def test_hetero_partition_roundtrip(tmp_path):
         assert seen_edges[et] == want, et
     assert seen_rows["u"] == set(range(n_u))
     assert seen_rows["v"] == set(range(n_v))
+
+
+def test_feature_pickle_roundtrip():
+    """CPU Feature survives pickling (the loader worker/IPC path)."""
+    import pickle
+
+    from glt_amd.data import Feature
+
+    feats = torch.arange(20, dtype=torch.float32).unsqueeze(1).repeat(1, 3)
+    id2index = torch.flip(torch.arange(20), [0])  # reordered store
+    f = Feature(feats[id2index.argsort()][id2index], with_gpu=False,
+                id2index=id2index)
+    # id2index indirection resolves global ids
+    got = f[torch.tensor([3, 17])]
+    f2 = pickle.loads(pickle.dumps(f))
+    assert torch.equal(f2[torch.tensor([3, 17])], got)
+    assert torch.equal(f2[torch.tensor([0])], f[torch.tensor([0])])
