@@ -267,3 +267,25 @@ def test_feature_pickle_roundtrip():
     f2 = pickle.loads(pickle.dumps(f))
     assert torch.equal(f2[torch.tensor([3, 17])], got)
     assert torch.equal(f2[torch.tensor([0])], f[torch.tensor([0])])
+
+
+def test_tracing_noop_without_gpu():
+    """trace_region must be a safe no-op on CPU-only boxes (and still
+    propagate exceptions from the wrapped block)."""
+    from glt_amd.utils.tracing import range_pop, range_push, trace_region
+
+    range_push("x")
+    range_pop()
+    with trace_region("phase"):
+        v = 41 + 1
+    assert v == 42
+    import pytest as _pytest
+    with _pytest.raises(ValueError):
+        with trace_region("boom"):
+            raise ValueError("boom")
+
+
+def test_exit_status_guard():
+    from glt_amd.utils.exit_status import python_exit_status
+
+    assert python_exit_status() is False  # interpreter still alive
