@@ -88,3 +88,32 @@ def test_random_node_split(ring_graph):
     assert ds.train_idx.numel() == 28
     all_ids = torch.cat([ds.train_idx, ds.val_idx, ds.test_idx])
     assert set(all_ids.tolist()) == set(range(40))
+
+
+def test_index_select_nested():
+    from glt_amd.utils.common import index_select
+
+    assert index_select(None, torch.tensor([0])) is None
+    d = {"a": torch.arange(5), "b": {"c": torch.arange(10, 15)}}
+    out = index_select(d, torch.tensor([1, 3]))
+    assert out["a"].tolist() == [1, 3]
+    assert out["b"]["c"].tolist() == [11, 13]
+
+
+def test_format_hetero_sampler_output():
+    from glt_amd.sampler import HeteroSamplerOutput
+    from glt_amd.utils import format_hetero_sampler_output
+
+    et = ("u", "e", "w")  # 'w' never appears as a node dict key
+    out = HeteroSamplerOutput(
+        node={"u": torch.tensor([1])}, row={et: torch.tensor([0])},
+        col={et: torch.tensor([0])})
+    fixed = format_hetero_sampler_output(out)
+    assert "w" in fixed.node and fixed.node["w"].numel() == 0
+
+
+def test_assign_device():
+    from glt_amd.utils.common import assign_device
+
+    d = assign_device(0)
+    assert d.type in ("cuda", "cpu")  # cpu fallback without a GPU
