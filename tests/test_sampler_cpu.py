@@ -197,3 +197,28 @@ def test_deferred_gating(ring_graph):
     # edge-capacity gate arithmetic
     s2 = NeighborSampler(g, [15, 10, 5])
     assert s2._edge_cap(1024) == 1024 * 15 * (1 + 10 + 10 * 5)
+
+
+def test_sampling_reproducible_under_seed(ring_graph):
+    """seed_everything makes full multi-hop sampling runs bitwise
+    reproducible (SeedManager base seed + call counter)."""
+    from glt_amd.data import Graph, Topology
+    from glt_amd.sampler import NeighborSampler, NodeSamplerInput
+
+    topo = Topology(ring_graph["edge_index"], num_nodes=40)
+    g = Graph(topo, mode="CPU")
+
+    def run():
+        glt_amd.seed_everything(123)
+        s = NeighborSampler(g, [1, 1])
+        outs = []
+        for start in (0, 8, 16):
+            seeds = torch.arange(start, start + 8)
+            outs.append(s.sample_from_nodes(NodeSamplerInput(node=seeds)))
+        return outs
+
+    a, b = run(), run()
+    for x, y in zip(a, b):
+        assert torch.equal(x.node, y.node)
+        assert torch.equal(x.row, y.row)
+        assert torch.equal(x.col, y.col)
