@@ -309,7 +309,11 @@ def _load_graph_dir(sub):
         return None
     rows = torch.load(os.path.join(sub, "rows.pt"), weights_only=False)
     cols = torch.load(os.path.join(sub, "cols.pt"), weights_only=False)
-    eids = torch.load(os.path.join(sub, "eids.pt"), weights_only=False)
+    epath = os.path.join(sub, "eids.pt")
+    # save_graph_cache writes eids only with_edge_feat=True; synthesize
+    # positional ids otherwise
+    eids = torch.load(epath, weights_only=False) \
+        if os.path.exists(epath) else torch.arange(rows.numel())
     wpath = os.path.join(sub, "weights.pt")
     weights = torch.load(wpath, weights_only=False) \
         if os.path.exists(wpath) else None

@@ -289,3 +289,35 @@ def test_exit_status_guard():
     from glt_amd.utils.exit_status import python_exit_status
 
     assert python_exit_status() is False  # interpreter still alive
+
+
+def test_graph_caching_load(tmp_path, ring_graph):
+    """graph_caching=True: every rank loads the WHOLE topology from the
+    root-level cache while features stay partitioned (reference
+    partition/base.py:93-118 mode).  Covers the eids-less default cache."""
+    from glt_amd.distributed import DistDataset
+    from glt_amd.partition import RandomPartitioner, load_partition
+    from glt_amd.partition.base import save_graph_cache
+
+    glt_amd.seed_everything(9)
+    feats = ring_graph["feats"]
+    p = RandomPartitioner(str(tmp_path), num_parts=2, num_nodes=40,
+                          edge_index=ring_graph["edge_index"],
+                          node_feat=feats)
+    p.partition()
+    parts = [load_partition(str(tmp_path), i)[1] for i in range(2)]
+    save_graph_cache(str(tmp_path), parts)
+
+    ds = DistDataset()
+    ds.load(str(tmp_path), 0, graph_mode="CPU", feature_with_gpu=False,
+            graph_caching=True)
+    # the rank sees the FULL ring topology...
+    assert ds.graph.num_edges == 80
+    rows, cols, _ = ds.graph.topo.to_coo()
+    assert ((cols - rows) % 40 <= 2).all()
+    # ...but only its own feature shard
+    local = (ds.node_pb.book == 0).sum().item()
+    assert 0 < local < 40
+    ids = torch.nonzero(ds.node_pb.book == 0).flatten()
+    got = ds.node_features[ids]
+    assert torch.equal(got, feats[ids])
