@@ -308,9 +308,13 @@ def test_graph_caching_load(tmp_path, ring_graph):
     parts = [load_partition(str(tmp_path), i)[1] for i in range(2)]
     save_graph_cache(str(tmp_path), parts)
 
+    labels = ring_graph["labels"]
+    torch.save(labels, str(tmp_path / "labels.pt"))
     ds = DistDataset()
     ds.load(str(tmp_path), 0, graph_mode="CPU", feature_with_gpu=False,
-            graph_caching=True)
+            graph_caching=True,
+            whole_node_label_file=str(tmp_path / "labels.pt"))
+    assert torch.equal(ds.node_labels, labels)
     # the rank sees the FULL ring topology...
     assert ds.graph.num_edges == 80
     rows, cols, _ = ds.graph.topo.to_coo()
