@@ -325,3 +325,28 @@ def test_graph_caching_load(tmp_path, ring_graph):
     ids = torch.nonzero(ds.node_pb.book == 0).flatten()
     got = ds.node_features[ids]
     assert torch.equal(got, feats[ids])
+
+
+def test_empty_and_degenerate_inputs(ring_graph):
+    """Degenerate inputs must not crash: empty seed tensor, isolated
+    nodes (degree 0), single-node batches."""
+    from glt_amd.data import Graph, Topology
+    from glt_amd.sampler import NeighborSampler, NodeSamplerInput
+
+    # graph with isolated node 41 appended
+    ei = ring_graph["edge_index"]
+    topo = Topology(ei, num_nodes=42)
+    g = Graph(topo, mode="CPU")
+    s = NeighborSampler(g, [2, 2])
+    # isolated seed: present in output, no edges from it at hop 1
+    out = s.sample_from_nodes(NodeSamplerInput(node=torch.tensor([41])))
+    assert out.node[0].item() == 41
+    assert out.num_sampled_edges[0] == 0
+    # empty seeds
+    out = s.sample_from_nodes(
+        NodeSamplerInput(node=torch.empty(0, dtype=torch.long)))
+    assert out.node.numel() == 0 and out.row.numel() == 0
+    # single repeated seed dedups to one
+    out = s.sample_from_nodes(
+        NodeSamplerInput(node=torch.tensor([7, 7, 7])))
+    assert out.batch.tolist() == [7]
