@@ -350,3 +350,43 @@ def test_empty_and_degenerate_inputs(ring_graph):
     out = s.sample_from_nodes(
         NodeSamplerInput(node=torch.tensor([7, 7, 7])))
     assert out.batch.tolist() == [7]
+
+
+def test_concurrent_event_loop():
+    """Thread-hosted asyncio loop: bounded concurrency, run/add task,
+    clean shutdown (the DistNeighborSampler execution substrate)."""
+    import asyncio
+    import threading
+
+    from glt_amd.distributed.event_loop import ConcurrentEventLoop
+
+    ev = ConcurrentEventLoop(concurrency=2)
+    ev.start_loop()
+    peak = [0]
+    live = [0]
+    lock = threading.Lock()
+
+    async def work(i):
+        with lock:
+            live[0] += 1
+            peak[0] = max(peak[0], live[0])
+        await asyncio.sleep(0.02)
+        with lock:
+            live[0] -= 1
+        return i * i
+
+    futs = [ev.add_task(work(i)) for i in range(8)]
+    assert sorted(f.result(timeout=10) for f in futs) == \
+        [i * i for i in range(8)]
+    assert peak[0] <= 2  # semaphore bound respected
+    assert ev.run_task(work(9)) == 81
+    ev.shutdown_loop()
+
+
+def test_rpc_partition_router():
+    from glt_amd.distributed.rpc import RpcDataPartitionRouter
+
+    r = RpcDataPartitionRouter([["w0", "w1"], ["w2"]])
+    assert [r.get_to_worker(0) for _ in range(4)] == \
+        ["w0", "w1", "w0", "w1"]
+    assert [r.get_to_worker(1) for _ in range(3)] == ["w2"] * 3
