@@ -390,3 +390,29 @@ def test_rpc_partition_router():
     assert [r.get_to_worker(0) for _ in range(4)] == \
         ["w0", "w1", "w0", "w1"]
     assert [r.get_to_worker(1) for _ in range(3)] == ["w2"] * 3
+
+
+def test_wrap_torch_future():
+    """torch.futures bridge into the sampler's asyncio loop (value and
+    exception paths)."""
+    from glt_amd.distributed.event_loop import (ConcurrentEventLoop,
+                                                wrap_torch_future)
+
+    ev = ConcurrentEventLoop(2)
+    ev.start_loop()
+
+    async def use(tf):
+        return await wrap_torch_future(ev.loop, tf)
+
+    tf = torch.futures.Future()
+    fut = ev.add_task(use(tf))
+    tf.set_result(42)
+    assert fut.result(timeout=10) == 42
+
+    tf2 = torch.futures.Future()
+    fut2 = ev.add_task(use(tf2))
+    tf2.set_exception(RuntimeError("boom"))
+    import pytest as _pytest
+    with _pytest.raises(Exception, match="boom"):
+        fut2.result(timeout=10)
+    ev.shutdown_loop()
