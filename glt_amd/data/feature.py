@@ -124,10 +124,23 @@ class Feature:
         self._device_rows = hot
 
     # -- lookups ------------------------------------------------------------
+    def _id2index_on(self, device) -> torch.Tensor:
+        """Device-resident id2index, cached per device — converting per
+        lookup would re-upload the full [max_id+1] int64 map every batch
+        (hundreds of MB at papers100M scale)."""
+        key = (device.type, device.index)
+        cache = getattr(self, "_id2index_cache", None)
+        if cache is None:
+            cache = self._id2index_cache = {}
+        t = cache.get(key)
+        if t is None:
+            t = cache[key] = self.id2index.to(device)
+        return t
+
     def __getitem__(self, ids: torch.Tensor) -> torch.Tensor:
         rows = ids.long()
         if self.id2index is not None:
-            rows = self.id2index.to(rows.device)[rows]
+            rows = self._id2index_on(rows.device)[rows]
         if self.with_gpu:
             self.lazy_init()
             dev = self.device if self.device is not None else \

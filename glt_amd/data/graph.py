@@ -15,6 +15,22 @@ import torch
 from ..utils.topo import coo_to_csc, coo_to_csr
 
 
+def _rows_sorted(indptr: torch.Tensor, indices: torch.Tensor) -> bool:
+    """O(E) vectorized check that `indices` is ascending within each CSR
+    row (descents are allowed exactly at row boundaries)."""
+    e = indices.numel()
+    if e < 2:
+        return True
+    descent = indices[1:] < indices[:-1]
+    if not bool(descent.any()):
+        return True
+    boundary = torch.zeros(e, dtype=torch.bool, device=indices.device)
+    starts = indptr[1:-1]
+    starts = starts[(starts > 0) & (starts < e)]
+    boundary[starts] = True
+    return not bool((descent & ~boundary[1:]).any())
+
+
 class Topology:
     """Layout-normalizing CSR/CSC/COO container.
 
@@ -56,6 +72,15 @@ class Topology:
             if eids is None:
                 eids = torch.arange(indices.numel(), dtype=torch.long)
             ew = edge_weights
+            if input_layout == layout and not _rows_sorted(indptr, indices):
+                # the negative samplers binary-search within rows
+                # (csrc/hip/hip_sampler.hip d_edge_in_csr + the CPU twin):
+                # enforce the per-row sorted-column invariant on passthrough
+                # input too, not just the COO path
+                from ..utils.topo import sort_csr_indices
+
+                indices, eids, ew = sort_csr_indices(indptr, indices,
+                                                     eids, ew)
             if input_layout != layout:
                 # convert via COO round trip
                 num_rows = indptr.numel() - 1

@@ -322,8 +322,7 @@ class DistNeighborSampler:
         s = self.sampler
         neg = inputs.neg_sampling
         if isinstance(self.data.graph, dict):
-            # delegate to local hetero link sampling then collect
-            out = s.sample_from_edges(inputs)
+            out = await self._hetero_sample_from_edges(inputs)
             return await self._collect(out)
         row = inputs.row.to(s._sample_device)
         col = inputs.col.to(s._sample_device)
@@ -374,6 +373,78 @@ class DistNeighborSampler:
                 if inputs.label is not None else None}
         return await self._collect(out)
 
+    async def _hetero_sample_from_edges(self, inputs: EdgeSamplerInput):
+        """Hetero link sampling with the full cross-partition fan-out.
+
+        Mirrors the local NeighborSampler._hetero_sample_from_edges but
+        runs the multi-hop expansion through _hetero_multihop so every hop
+        fans out over partitions via _sample_one_hop RPC (parity:
+        reference dist_neighbor_sampler.py:400-476 — the local-only
+        delegate would silently drop remote neighborhoods).  Negative
+        sampling still draws against the local partition's edge set
+        (approximate strict check, as in the homo branch above).
+        """
+        from ..sampler.neighbor_sampler import _relabel
+
+        s = self.sampler
+        etype = inputs.input_type
+        assert etype is not None
+        src_t, _, dst_t = etype
+        row = inputs.row.to(s._sample_device)
+        col = inputs.col.to(s._sample_device)
+        num_pos = row.numel()
+        neg = inputs.neg_sampling
+        neg_dst = None
+        from .. import _C
+
+        if neg is not None:
+            g = s.graph[etype]
+            amount = (int(num_pos * neg.amount) if neg.is_binary()
+                      else num_pos * int(neg.amount))
+            neg_pair = _C.sample_negative(
+                g.indptr, g.indices, g.num_nodes, amount, 5, True)
+            neg_dst = neg_pair[1]
+        if src_t == dst_t:
+            seeds = {src_t: torch.cat([row, col] +
+                                      ([neg_dst] if neg_dst is not None
+                                       else []))}
+        else:
+            dst_seeds = torch.cat([col] + ([neg_dst]
+                                           if neg_dst is not None else []))
+            seeds = {src_t: row, dst_t: dst_seeds}
+        out = await self._hetero_multihop(
+            seeds, metadata={"input_type": etype})
+        src_local = _relabel(out.node[src_t], row)
+        dst_local = _relabel(out.node[dst_t], col)
+        if neg is not None and neg.is_triplet():
+            out.metadata.update({
+                "src_index": src_local,
+                "dst_pos_index": dst_local,
+                "dst_neg_index": _relabel(out.node[dst_t],
+                                          neg_dst).view(num_pos, -1),
+            })
+        else:
+            if neg_dst is not None:
+                neg_src = row.repeat(
+                    (neg_dst.numel() + num_pos - 1) // num_pos)[
+                        : neg_dst.numel()]
+                eli = torch.stack([
+                    torch.cat([src_local,
+                               _relabel(out.node[src_t], neg_src)]),
+                    torch.cat([dst_local,
+                               _relabel(out.node[dst_t], neg_dst)]),
+                ])
+                label = torch.cat([
+                    torch.ones(num_pos, device=s.device),
+                    torch.zeros(neg_dst.numel(), device=s.device)])
+            else:
+                eli = torch.stack([src_local, dst_local])
+                label = (inputs.label.to(s.device)
+                         if inputs.label is not None else None)
+            out.metadata.update({"edge_label_index": eli,
+                                 "edge_label": label})
+        return out
+
     async def _subgraph(self, inputs: NodeSamplerInput):
         out = self.sampler.subgraph(inputs)
         return await self._collect(out)
@@ -413,7 +484,11 @@ class DistNeighborSampler:
                                      edge_attr=ea or None)
 
     async def _collect_one(self, loop, kind, ids, type_key):
-        if self.use_all2all:
+        # all2all only for the homo path: hetero collection gates each
+        # (kind, type) on per-rank conditions (nodes.numel(), has()), so
+        # ranks could issue different numbers of all_to_all_single calls
+        # and deadlock the RCCL group — force the RPC pull path there.
+        if self.use_all2all and type_key is None and kind == "node":
             import torch.distributed as dist
 
             if dist.is_initialized() and \

@@ -133,10 +133,14 @@ class GATConv(nn.Module):
         nn.init.xavier_uniform_(self.att_src)
         nn.init.xavier_uniform_(self.att_dst)
 
-    def forward(self, x, edge_index, num_target: int = None):
+    def forward(self, x, edge_index, num_target: int = None,
+                sorted_by_target: bool = True):
         """x: [n, F] or a (x_target, x_source) tuple for bipartite edge
         sets (hetero relations); edge_index[0] indexes the target side,
-        edge_index[1] the source side."""
+        edge_index[1] the source side.  sorted_by_target: set False for
+        edge sets not sorted by edge_index[0] — the fused segment kernel
+        requires ascending targets; the scatter-softmax fallback does
+        not."""
         if isinstance(x, tuple):
             x_tgt, x_src = x
             nt = num_target if num_target is not None else x_tgt.size(0)
@@ -148,15 +152,18 @@ class GATConv(nn.Module):
             nt = num_target if num_target is not None else x.size(0)
             h_tgt = h_src = self.lin(x).view(x.size(0), self.heads,
                                              self.out_channels)
-        return self.attend(h_tgt, h_src, edge_index, nt)
+        return self.attend(h_tgt, h_src, edge_index, nt,
+                           sorted_by_target=sorted_by_target)
 
-    def attend(self, h_tgt, h_src, edge_index, nt):
+    def attend(self, h_tgt, h_src, edge_index, nt,
+               sorted_by_target: bool = True):
         """Attention + aggregation over pre-projected features
         [n, heads, C] (lets HeteroConv batch the projections of all
         relations sharing a node type into one GEMM)."""
         tgt, src = edge_index[0], edge_index[1]
         h = h_src
-        if (getattr(self, "use_fused", True) and h_src.is_cuda
+        if (sorted_by_target and getattr(self, "use_fused", True)
+                and h_src.is_cuda
                 and h_src.dtype == torch.float32
                 and self.out_channels <= 128
                 and not (self.training and self.dropout > 0)):

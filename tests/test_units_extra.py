@@ -416,3 +416,45 @@ def test_wrap_torch_future():
     with _pytest.raises(Exception, match="boom"):
         fut2.result(timeout=10)
     ev.shutdown_loop()
+
+
+def test_topology_csr_passthrough_sorts_rows():
+    """CSR passthrough input with unsorted columns must be row-sorted
+    (the negative sampler binary-searches within rows)."""
+    import torch
+
+    from glt_amd.data import Topology
+
+    indptr = torch.tensor([0, 3, 5, 5, 6])
+    indices = torch.tensor([2, 0, 1, 3, 1, 0])
+    ew = torch.tensor([2.0, 0.0, 1.0, 3.0, 1.0, 0.0])
+    topo = Topology((indptr, indices), edge_weights=ew,
+                    input_layout="CSR", layout="CSR")
+    assert topo.indices.tolist() == [0, 1, 2, 1, 3, 0]
+    # aux arrays permuted in lockstep (weight == original column here)
+    assert topo.edge_weights.tolist() == [0.0, 1.0, 2.0, 1.0, 3.0, 0.0]
+    # already-sorted input passes through untouched (no copy/perm)
+    t2 = Topology((topo.indptr, topo.indices), input_layout="CSR",
+                  layout="CSR")
+    assert t2.indices.data_ptr() == topo.indices.data_ptr()
+
+
+def test_gatconv_unsorted_edges_opt_out():
+    """sorted_by_target=False must give the same result as a sorted edge
+    list through the fused path's fallback."""
+    import torch
+
+    from glt_amd.models.layers import GATConv
+
+    torch.manual_seed(0)
+    conv = GATConv(8, 4, heads=2)
+    n, e = 10, 30
+    x = torch.randn(n, 8)
+    tgt = torch.sort(torch.randint(0, 6, (e,))).values
+    src = torch.randint(0, n, (e,))
+    ei = torch.stack([tgt, src])
+    out_sorted = conv(x, ei, num_target=6)
+    perm = torch.randperm(e)
+    ei_shuf = ei[:, perm]
+    out_shuf = conv(x, ei_shuf, num_target=6, sorted_by_target=False)
+    assert torch.allclose(out_sorted, out_shuf, atol=1e-5)
