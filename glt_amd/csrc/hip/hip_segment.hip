@@ -8,6 +8,12 @@
  *
  * Replaces the 4-kernel torch sequence (bincount + index_select +
  * index_add_ + div) in SAGEConv (glt_amd/models/layers.py).
+ *
+ * dtype: templated over fp32 and bf16 inputs with fp32 accumulation —
+ * the kernels are HBM-bound (see profiles/), so bf16 halves the roofline
+ * traffic.  Backward always accumulates into an fp32 arena (bf16 atomics
+ * lose too much precision for degree-weighted sums); the python wrapper
+ * casts once at the end.
  */
 #include "hip_common.h"
 #include "../include/common.h"
@@ -18,11 +24,12 @@ namespace {
 
 // out[t, :] = mean_{e in [off[t], off[t+1])} x[col[e], :]
 // One wave per target row; lanes cover the feature dim.
-__global__ void seg_mean_fwd_kernel(const float* __restrict__ x,
+template <typename scalar_t>
+__global__ void seg_mean_fwd_kernel(const scalar_t* __restrict__ x,
                                     const int64_t* __restrict__ col,
                                     const int64_t* __restrict__ off,
                                     int64_t n_tgt, int64_t feat,
-                                    float* __restrict__ out) {
+                                    scalar_t* __restrict__ out) {
   const int lane = threadIdx.x & (kWave - 1);
   const int64_t wave =
       (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) / kWave;
@@ -32,14 +39,16 @@ __global__ void seg_mean_fwd_kernel(const float* __restrict__ x,
     const float inv = e > s ? 1.0f / (float)(e - s) : 0.0f;
     for (int64_t f = lane; f < feat; f += kWave) {
       float acc = 0.f;
-      for (int64_t k = s; k < e; ++k) acc += x[col[k] * feat + f];
-      out[t * feat + f] = acc * inv;
+      for (int64_t k = s; k < e; ++k)
+        acc += static_cast<float>(x[col[k] * feat + f]);
+      out[t * feat + f] = static_cast<scalar_t>(acc * inv);
     }
   }
 }
 
-// dx[col[e], :] += dy[t, :] / deg(t)
-__global__ void seg_mean_bwd_kernel(const float* __restrict__ dy,
+// dx[col[e], :] += dy[t, :] / deg(t)  (dx is always fp32)
+template <typename scalar_t>
+__global__ void seg_mean_bwd_kernel(const scalar_t* __restrict__ dy,
                                     const int64_t* __restrict__ col,
                                     const int64_t* __restrict__ off,
                                     int64_t n_tgt, int64_t feat,
@@ -55,7 +64,8 @@ __global__ void seg_mean_bwd_kernel(const float* __restrict__ dy,
     for (int64_t k = s; k < e; ++k) {
       const int64_t c = col[k];
       for (int64_t f = lane; f < feat; f += kWave) {
-        atomicAdd(&dx[c * feat + f], dy[t * feat + f] * inv);
+        atomicAdd(&dx[c * feat + f],
+                  static_cast<float>(dy[t * feat + f]) * inv);
       }
     }
   }
@@ -66,11 +76,12 @@ __global__ void seg_mean_bwd_kernel(const float* __restrict__ dy,
 // over [n, F] per layer, ~260 us/step in the flagship profile) and its
 // launches; valid for the homo SAGE path where targets are the row prefix
 // of x (the sampler's local-id invariant).
-__global__ void seg_mean_cat_fwd_kernel(const float* __restrict__ x,
+template <typename scalar_t>
+__global__ void seg_mean_cat_fwd_kernel(const scalar_t* __restrict__ x,
                                         const int64_t* __restrict__ col,
                                         const int64_t* __restrict__ off,
                                         int64_t n_tgt, int64_t feat,
-                                        float* __restrict__ out) {
+                                        scalar_t* __restrict__ out) {
   const int lane = threadIdx.x & (kWave - 1);
   const int64_t wave =
       (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) / kWave;
@@ -81,15 +92,17 @@ __global__ void seg_mean_cat_fwd_kernel(const float* __restrict__ x,
     const float inv = e > s ? 1.0f / (float)(e - s) : 0.0f;
     for (int64_t f = lane; f < feat; f += kWave) {
       float acc = 0.f;
-      for (int64_t k = s; k < e; ++k) acc += x[col[k] * feat + f];
-      out[t * ostride + f] = acc * inv;
+      for (int64_t k = s; k < e; ++k)
+        acc += static_cast<float>(x[col[k] * feat + f]);
+      out[t * ostride + f] = static_cast<scalar_t>(acc * inv);
       out[t * ostride + feat + f] = x[t * feat + f];
     }
   }
 }
 
 // dx[t, :] += dy[t, F:];  dx[col[e], :] += dy[t, :F] / deg(t)
-__global__ void seg_mean_cat_bwd_kernel(const float* __restrict__ dy,
+template <typename scalar_t>
+__global__ void seg_mean_cat_bwd_kernel(const scalar_t* __restrict__ dy,
                                         const int64_t* __restrict__ col,
                                         const int64_t* __restrict__ off,
                                         int64_t n_tgt, int64_t feat,
@@ -102,13 +115,15 @@ __global__ void seg_mean_cat_bwd_kernel(const float* __restrict__ dy,
   for (int64_t t = wave; t < n_tgt; t += n_waves) {
     const int64_t s = off[t], e = off[t + 1];
     for (int64_t f = lane; f < feat; f += kWave)
-      atomicAdd(&dx[t * feat + f], dy[t * ostride + feat + f]);
+      atomicAdd(&dx[t * feat + f],
+                static_cast<float>(dy[t * ostride + feat + f]));
     if (e <= s) continue;
     const float inv = 1.0f / (float)(e - s);
     for (int64_t k = s; k < e; ++k) {
       const int64_t c = col[k];
       for (int64_t f = lane; f < feat; f += kWave)
-        atomicAdd(&dx[c * feat + f], dy[t * ostride + f] * inv);
+        atomicAdd(&dx[c * feat + f],
+                  static_cast<float>(dy[t * ostride + f]) * inv);
     }
   }
 }
@@ -119,23 +134,41 @@ int wave_grid(int64_t rows) {
                                 kMaxBlocks);
 }
 
+#define GLT_DISPATCH_SEG(TYPE, NAME, ...)                          \
+  [&] {                                                            \
+    switch (TYPE) {                                                \
+      case torch::kFloat32: {                                      \
+        using scalar_t = float;                                    \
+        return __VA_ARGS__();                                      \
+      }                                                            \
+      case torch::kBFloat16: {                                     \
+        using scalar_t = c10::BFloat16;                            \
+        return __VA_ARGS__();                                      \
+      }                                                            \
+      default:                                                     \
+        TORCH_CHECK(false, NAME ": unsupported dtype (fp32/bf16)");\
+    }                                                              \
+  }()
+
 }  // namespace
 
 torch::Tensor hip_segment_mean_fwd(const torch::Tensor& x,
                                    const torch::Tensor& col,
                                    const torch::Tensor& offsets,
                                    int64_t n_tgt) {
-  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kFloat32 &&
-                  x.is_contiguous(),
-              "segment_mean: x must be contiguous fp32 on device");
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(),
+              "segment_mean: x must be contiguous on device");
   const int64_t feat = x.size(1);
   auto out = torch::empty({n_tgt, feat}, x.options());
   if (n_tgt > 0) {
-    hipLaunchKernelGGL(seg_mean_fwd_kernel, dim3(wave_grid(n_tgt)),
-                       dim3(kBlock), 0, current_stream(),
-                       x.data_ptr<float>(), col.data_ptr<int64_t>(),
-                       offsets.data_ptr<int64_t>(), n_tgt, feat,
-                       out.data_ptr<float>());
+    GLT_DISPATCH_SEG(x.scalar_type(), "segment_mean_fwd", [&] {
+      hipLaunchKernelGGL(seg_mean_fwd_kernel<scalar_t>,
+                         dim3(wave_grid(n_tgt)), dim3(kBlock), 0,
+                         current_stream(), x.data_ptr<scalar_t>(),
+                         col.data_ptr<int64_t>(),
+                         offsets.data_ptr<int64_t>(), n_tgt, feat,
+                         out.data_ptr<scalar_t>());
+    });
   }
   return out;
 }
@@ -144,37 +177,44 @@ torch::Tensor hip_segment_mean_cat_fwd(const torch::Tensor& x,
                                        const torch::Tensor& col,
                                        const torch::Tensor& offsets,
                                        int64_t n_tgt) {
-  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kFloat32 &&
-                  x.is_contiguous(),
-              "segment_mean_cat: x must be contiguous fp32 on device");
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(),
+              "segment_mean_cat: x must be contiguous on device");
   TORCH_CHECK(x.size(0) >= n_tgt,
               "segment_mean_cat: targets must be a row prefix of x");
   const int64_t feat = x.size(1);
   auto out = torch::empty({n_tgt, 2 * feat}, x.options());
   if (n_tgt > 0) {
-    hipLaunchKernelGGL(seg_mean_cat_fwd_kernel, dim3(wave_grid(n_tgt)),
-                       dim3(kBlock), 0, current_stream(),
-                       x.data_ptr<float>(), col.data_ptr<int64_t>(),
-                       offsets.data_ptr<int64_t>(), n_tgt, feat,
-                       out.data_ptr<float>());
+    GLT_DISPATCH_SEG(x.scalar_type(), "segment_mean_cat_fwd", [&] {
+      hipLaunchKernelGGL(seg_mean_cat_fwd_kernel<scalar_t>,
+                         dim3(wave_grid(n_tgt)), dim3(kBlock), 0,
+                         current_stream(), x.data_ptr<scalar_t>(),
+                         col.data_ptr<int64_t>(),
+                         offsets.data_ptr<int64_t>(), n_tgt, feat,
+                         out.data_ptr<scalar_t>());
+    });
   }
   return out;
 }
 
+// Returns fp32 regardless of dy dtype (fp32 atomic accumulation arena);
+// the python wrapper casts to dy.dtype once.
 torch::Tensor hip_segment_mean_cat_bwd(const torch::Tensor& dy,
                                        const torch::Tensor& col,
                                        const torch::Tensor& offsets,
                                        int64_t n_src) {
   const int64_t n_tgt = dy.size(0);
   const int64_t feat = dy.size(1) / 2;
-  auto dx = torch::zeros({n_src, feat}, dy.options());
+  auto dx = torch::zeros({n_src, feat}, dy.options().dtype(torch::kFloat32));
   if (n_tgt > 0) {
-    hipLaunchKernelGGL(seg_mean_cat_bwd_kernel, dim3(wave_grid(n_tgt)),
-                       dim3(kBlock), 0, current_stream(),
-                       dy.contiguous().data_ptr<float>(),
-                       col.data_ptr<int64_t>(),
-                       offsets.data_ptr<int64_t>(), n_tgt, feat,
-                       dx.data_ptr<float>());
+    auto dyc = dy.contiguous();
+    GLT_DISPATCH_SEG(dy.scalar_type(), "segment_mean_cat_bwd", [&] {
+      hipLaunchKernelGGL(seg_mean_cat_bwd_kernel<scalar_t>,
+                         dim3(wave_grid(n_tgt)), dim3(kBlock), 0,
+                         current_stream(), dyc.data_ptr<scalar_t>(),
+                         col.data_ptr<int64_t>(),
+                         offsets.data_ptr<int64_t>(), n_tgt, feat,
+                         dx.data_ptr<float>());
+    });
   }
   return dx;
 }
@@ -185,14 +225,17 @@ torch::Tensor hip_segment_mean_bwd(const torch::Tensor& dy,
                                    int64_t n_src) {
   const int64_t n_tgt = dy.size(0);
   const int64_t feat = dy.size(1);
-  auto dx = torch::zeros({n_src, feat}, dy.options());
+  auto dx = torch::zeros({n_src, feat}, dy.options().dtype(torch::kFloat32));
   if (n_tgt > 0) {
-    hipLaunchKernelGGL(seg_mean_bwd_kernel, dim3(wave_grid(n_tgt)),
-                       dim3(kBlock), 0, current_stream(),
-                       dy.contiguous().data_ptr<float>(),
-                       col.data_ptr<int64_t>(),
-                       offsets.data_ptr<int64_t>(), n_tgt, feat,
-                       dx.data_ptr<float>());
+    auto dyc = dy.contiguous();
+    GLT_DISPATCH_SEG(dy.scalar_type(), "segment_mean_bwd", [&] {
+      hipLaunchKernelGGL(seg_mean_bwd_kernel<scalar_t>,
+                         dim3(wave_grid(n_tgt)), dim3(kBlock), 0,
+                         current_stream(), dyc.data_ptr<scalar_t>(),
+                         col.data_ptr<int64_t>(),
+                         offsets.data_ptr<int64_t>(), n_tgt, feat,
+                         dx.data_ptr<float>());
+    });
   }
   return dx;
 }

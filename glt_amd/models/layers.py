@@ -47,8 +47,8 @@ class SAGEConv(nn.Module):
             x_tgt, x_src = x
             n = num_target if num_target is not None else x_tgt.size(0)
             tgt, src = edge_index[0], edge_index[1]
-            if x_src.is_cuda and x_src.dtype == torch.float32 and \
-                    sorted_by_target:
+            if x_src.is_cuda and sorted_by_target and \
+                    x_src.dtype in (torch.float32, torch.bfloat16):
                 from ..ops import segment_mean
 
                 agg = segment_mean(x_src, tgt, src, n)
@@ -60,12 +60,12 @@ class SAGEConv(nn.Module):
                 xin = torch.cat([agg, x_tgt[:n]], dim=1)
             else:
                 xin = agg
-            return self.lin(xin)
+            return self._project(xin, fuse_relu=False)
         n = num_target if num_target is not None else x.size(0)
         tgt, src = edge_index[0], edge_index[1]
-        from ..ops import mfma_linear, use_mfma_linear
 
-        if (x.is_cuda and x.dtype == torch.float32 and sorted_by_target):
+        if (x.is_cuda and sorted_by_target
+                and x.dtype in (torch.float32, torch.bfloat16)):
             # fused wave-per-row segment mean (glt_amd batches are sorted
             # by target local id by construction); with root_weight the
             # kernel also assembles [agg | x[:n]] in place of a dim-1 cat
@@ -83,6 +83,16 @@ class SAGEConv(nn.Module):
             agg = agg / _degree(tgt, n).unsqueeze(1).to(x.dtype)
             xin = torch.cat([agg, x[:n]], dim=1) if self.root_weight \
                 else agg
+        return self._project(xin, fuse_relu)
+
+    def _project(self, xin, fuse_relu: bool):
+        from ..ops import cast_linear, mfma_linear, use_mfma_linear
+
+        if xin.dtype != self.lin.weight.dtype:
+            # reduced-precision compute over fp32 master params (bf16
+            # batches from a bf16 feature store)
+            return cast_linear(xin, self.lin.weight, self.lin.bias,
+                               relu=fuse_relu)
         if use_mfma_linear(xin, self.lin.weight, relu=fuse_relu):
             return mfma_linear(xin, self.lin.weight, self.lin.bias,
                                relu=fuse_relu)

@@ -35,7 +35,10 @@ class _SegmentMean(torch.autograd.Function):
         from .. import _C
 
         col, offsets = ctx.saved_tensors
+        # kernel accumulates in an fp32 arena regardless of dy dtype
         dx = _C.segment_mean_bwd(dy.contiguous(), col, offsets, ctx.n_src)
+        if dx.dtype != dy.dtype:
+            dx = dx.to(dy.dtype)
         return dx, None, None, None
 
 
@@ -60,6 +63,8 @@ class _SegmentMeanCat(torch.autograd.Function):
         col, offsets = ctx.saved_tensors
         dx = _C.segment_mean_cat_bwd(dy.contiguous(), col, offsets,
                                      ctx.n_src)
+        if dx.dtype != dy.dtype:
+            dx = dx.to(dy.dtype)
         return dx, None, None, None
 
 

@@ -632,3 +632,129 @@ def test_deferred_out_of_range_seeds(ring_graph):
     rows = out.row.cpu().tolist()
     # only seeds 0 and 5 (local ids 0 and 3) contribute edges
     assert set(rows) <= {0, 3}
+
+
+def test_segment_mean_bf16_numerics():
+    """bf16 segment kernels (fp32 accumulation) vs the fp32 torch
+    reference — error must be at bf16 rounding scale, not accumulation
+    scale (long segments would drift if the kernel accumulated in
+    bf16)."""
+    from glt_amd.ops import segment_mean, segment_mean_cat
+
+    torch.manual_seed(0)
+    n_src, n_tgt, F, E = 4000, 64, 100, 64 * 300  # long segments
+    x32 = torch.randn(n_src, F, device="cuda")
+    x16 = x32.to(torch.bfloat16).requires_grad_(True)
+    tgt = torch.sort(torch.randint(0, n_tgt, (E,), device="cuda")).values
+    src = torch.randint(0, n_src, (E,), device="cuda")
+
+    out16 = segment_mean(x16, tgt, src, n_tgt)
+    assert out16.dtype == torch.bfloat16
+    # fp32 reference on the bf16-rounded inputs
+    xr = x16.detach().float()
+    agg = xr.new_zeros(n_tgt, F)
+    agg.index_add_(0, tgt, xr.index_select(0, src))
+    deg = torch.bincount(tgt, minlength=n_tgt).clamp(min=1)
+    ref = agg / deg.unsqueeze(1).float()
+    # one bf16 rounding of the fp32-accumulated mean
+    assert (out16.float() - ref).abs().max() < 8e-3, \
+        (out16.float() - ref).abs().max()
+
+    g = torch.randn(n_tgt, F, device="cuda").to(torch.bfloat16)
+    out16.backward(g)
+    assert x16.grad.dtype == torch.bfloat16
+    xr2 = xr.clone().requires_grad_(True)
+    agg2 = xr2.new_zeros(n_tgt, F)
+    agg2.index_add_(0, tgt, xr2.index_select(0, src))
+    (agg2 / deg.unsqueeze(1).float()).backward(g.float())
+    assert (x16.grad.float() - xr2.grad).abs().max() < 8e-3
+
+    # fused [agg | root] variant
+    x16b = x32.to(torch.bfloat16).requires_grad_(True)
+    outc = segment_mean_cat(x16b, tgt, src, n_tgt)
+    assert outc.dtype == torch.bfloat16
+    refc = torch.cat([ref, xr[:n_tgt]], dim=1)
+    assert (outc.float() - refc).abs().max() < 8e-3
+    gc = torch.randn_like(outc)
+    outc.backward(gc)
+    assert x16b.grad.dtype == torch.bfloat16
+
+
+def test_sageconv_bf16_matches_fp32():
+    """Full SAGEConv layer in bf16 (fused segment kernel + cast-linear
+    over fp32 master weights) vs the same layer in fp32."""
+    from glt_amd.models.layers import SAGEConv
+
+    torch.manual_seed(1)
+    conv = SAGEConv(64, 64).cuda()
+    n_src, n_tgt, E = 600, 100, 2400
+    x32 = torch.randn(n_src, 64, device="cuda", requires_grad=True)
+    tgt = torch.sort(torch.randint(0, n_tgt, (E,), device="cuda")).values
+    src = torch.randint(0, n_src, (E,), device="cuda")
+    ei = torch.stack([tgt, src])
+
+    out32 = conv(x32, ei, num_target=n_tgt)
+    x16 = x32.detach().to(torch.bfloat16).requires_grad_(True)
+    out16 = conv(x16, ei, num_target=n_tgt)
+    assert out16.dtype == torch.bfloat16
+    assert (out16.float() - out32).abs().max() < 0.15  # bf16 GEMM scale
+
+    # grads reach the fp32 master weights in fp32
+    out16.sum().backward()
+    assert conv.lin.weight.grad is not None
+    assert conv.lin.weight.grad.dtype == torch.float32
+    assert x16.grad.dtype == torch.bfloat16
+
+
+def test_end_to_end_learning_sbm_bf16():
+    """bf16 accuracy gate: the SBM community-recovery task must reach the
+    same >85% train accuracy as the fp32 pipeline (VERDICT round-1 #1)."""
+    from glt_amd import Dataset, NeighborLoader
+    from glt_amd.models import GraphSAGE
+
+    glt_amd.seed_everything(0)
+    n, k = 20_000, 10
+    comm = torch.randint(0, k, (n,))
+    e = n * 20
+    src = torch.randint(0, n, (e,))
+    same = torch.rand(e) < 0.9
+    dst = torch.randint(0, n, (e,))
+    order = torch.argsort(comm)
+    counts = torch.bincount(comm, minlength=k)
+    starts = torch.cat([torch.zeros(1, dtype=torch.long),
+                        torch.cumsum(counts, 0)[:-1]])
+    rnd = torch.rand(e)
+    same_dst = order[(starts[comm[src]] +
+                      (rnd * counts[comm[src]].float()).long().clamp(
+                          max=counts.max() - 1).clamp(min=0)) % n]
+    dst = torch.where(same, same_dst, dst)
+    feats = torch.nn.functional.one_hot(comm, k).float()
+    feats = feats + 0.5 * torch.randn(n, k)
+    ds = Dataset()
+    ds.init_graph(edge_index=torch.stack([src, dst]), graph_mode="CUDA",
+                  num_nodes=n, device=0)
+    ds.init_node_features(feats.to(torch.bfloat16), split_ratio=1.0,
+                          device=0)
+    ds.init_node_labels(comm.cuda())
+    dev = torch.device("cuda", 0)
+    loader = NeighborLoader(ds, [10, 5], input_nodes=torch.arange(n),
+                            batch_size=1024, shuffle=True, device=dev,
+                            to_device=dev, prefetch=2)
+    model = GraphSAGE(k, 64, 2, out_channels=k).to(dev)
+    opt = torch.optim.Adam(model.parameters(), lr=5e-3)
+    correct = total = 0
+    for epoch in range(3):
+        for data in loader:
+            assert data.x.dtype == torch.bfloat16
+            opt.zero_grad(set_to_none=True)
+            out = model(data.x, data.edge_index, data.num_sampled_nodes,
+                        data.num_sampled_edges)[:data.batch_size]
+            y = data.y[:data.batch_size]
+            loss = torch.nn.functional.cross_entropy(out.float(), y)
+            loss.backward()
+            opt.step()
+            if epoch == 2:
+                correct += int((out.argmax(-1) == y).sum())
+                total += y.numel()
+    acc = correct / max(total, 1)
+    assert acc > 0.85, f"bf16 end-to-end accuracy too low: {acc}"

@@ -458,3 +458,50 @@ def test_gatconv_unsorted_edges_opt_out():
     ei_shuf = ei[:, perm]
     out_shuf = conv(x, ei_shuf, num_target=6, sorted_by_target=False)
     assert torch.allclose(out_sorted, out_shuf, atol=1e-5)
+
+
+def test_cast_linear_cpu_grads():
+    """cast_linear: bf16 compute over fp32 master params; grads for the
+    fp32 leaves come back fp32 and match a plain fp32 linear at bf16
+    tolerance."""
+    import torch
+
+    from glt_amd.ops import cast_linear
+
+    torch.manual_seed(0)
+    w = torch.randn(16, 32, requires_grad=True)
+    b = torch.randn(16, requires_grad=True)
+    x32 = torch.randn(8, 32)
+    x16 = x32.to(torch.bfloat16).requires_grad_(True)
+    out = cast_linear(x16, w, b, relu=True)
+    assert out.dtype == torch.bfloat16
+    out.float().sum().backward()
+    assert w.grad is not None and w.grad.dtype == torch.float32
+    assert b.grad is not None and b.grad.dtype == torch.float32
+    assert x16.grad is not None and x16.grad.dtype == torch.bfloat16
+
+    # reference: fp32 path on the rounded input
+    w2 = w.detach().clone().requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+    ref = torch.relu(torch.nn.functional.linear(
+        x16.detach().float(), w2, b2))
+    assert (out.float() - ref).abs().max() < 0.12  # bf16 ulp at |out|~8
+    ref.sum().backward()
+    assert (w.grad - w2.grad).abs().max() < 0.1
+
+
+def test_sageconv_bf16_cpu_fallback():
+    """SAGEConv on CPU bf16 input: index_add fallback + cast_linear."""
+    import torch
+
+    from glt_amd.models.layers import SAGEConv
+
+    torch.manual_seed(0)
+    conv = SAGEConv(8, 8)
+    x = torch.randn(20, 8).to(torch.bfloat16).requires_grad_(True)
+    tgt = torch.sort(torch.randint(0, 10, (40,))).values
+    src = torch.randint(0, 20, (40,))
+    out = conv(x, torch.stack([tgt, src]), num_target=10)
+    assert out.dtype == torch.bfloat16
+    out.float().sum().backward()
+    assert conv.lin.weight.grad.dtype == torch.float32
