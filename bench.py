@@ -60,6 +60,12 @@ def parse_args():
                         "slower on ROCm 7.0 (hipGraph per-node overhead "
                         "~130us dominates this many-small-kernel step); "
                         "kept as a recorded experiment, default off")
+    p.add_argument("--graph-skew", type=float, default=0.0,
+                   help="power-law skew: >0 builds a Chung-Lu graph with "
+                        "lognormal(sigma=GRAPH_SKEW) expected degrees "
+                        "(mean held at 2E/N) instead of uniform "
+                        "Erdos-Renyi; 1.4 approximates ogbn-products' "
+                        "hub-heavy histogram (max deg ~35k)")
     p.add_argument("--feature-mode", type=str, default="replicated",
                    choices=["replicated", "xgmi-shard"],
                    help="xgmi-shard: features sharded across ranks' HBM, "
@@ -68,7 +74,15 @@ def parse_args():
 
 
 def build_synthetic(args, device, rank):
-    """ogbn-products-shaped uniform random graph, built on-device."""
+    """ogbn-products-shaped random graph, built on-device.
+
+    --graph-skew 0 (default): uniform Erdos-Renyi endpoints.
+    --graph-skew s>0: Chung-Lu — both endpoints drawn with probability
+    proportional to per-node lognormal(sigma=s) weights (inverse-CDF via
+    searchsorted), giving the hub-heavy power-law degree histogram of
+    real ogbn-products (hot-row gather contention + variable-degree
+    sampler rows that the uniform graph flatters away).
+    """
     import glt_amd
 
     glt_amd.seed_everything(args.seed + rank)
@@ -76,39 +90,42 @@ def build_synthetic(args, device, rank):
     gen_dev = device if device.type == "cuda" else torch.device("cpu")
     g = torch.Generator(device=gen_dev)
     g.manual_seed(args.seed)  # same graph on every rank
-    src = torch.randint(0, n, (e,), device=gen_dev, generator=g)
-    dst = torch.randint(0, n, (e,), device=gen_dev, generator=g)
+    if args.graph_skew > 0:
+        w = torch.randn(n, device=gen_dev, generator=g)
+        w = (w * args.graph_skew).exp()  # lognormal; mean set by normalizer
+        cdf = torch.cumsum(w, 0)
+        cdf /= cdf[-1].clone()
+        src = torch.searchsorted(
+            cdf, torch.rand(e, device=gen_dev, generator=g)).clamp_(0, n - 1)
+        dst = torch.searchsorted(
+            cdf, torch.rand(e, device=gen_dev, generator=g)).clamp_(0, n - 1)
+    else:
+        src = torch.randint(0, n, (e,), device=gen_dev, generator=g)
+        dst = torch.randint(0, n, (e,), device=gen_dev, generator=g)
     # undirected: both directions
     row = torch.cat([src, dst])
     col = torch.cat([dst, src])
-    # CSR build on device, then hand the Topology host/device tensors
-    perm = torch.argsort(row)
+    # CSR build on device, sorted by (row, col) so the per-row
+    # sorted-column invariant holds (the negative sampler binary-searches
+    # rows; fits int64 for n < 3e9 / sqrt)
+    perm = torch.argsort(row * n + col)
     row_s, col_s = row[perm], col[perm]
     counts = torch.bincount(row_s, minlength=n)
     indptr = torch.zeros(n + 1, dtype=torch.long, device=gen_dev)
     torch.cumsum(counts, 0, out=indptr[1:])
-    # sort indices within rows (needed by negative sampler only; cheap here
-    # via stable segmented sort: key = row * n + col would overflow for big
-    # graphs, so sort col within segments lazily -- uniform sampler does not
-    # need it; skip for bench)
-    eids = torch.arange(col_s.numel(), dtype=torch.long, device=gen_dev)
 
     from glt_amd.data import Graph, Topology
 
-    topo = Topology.__new__(Topology)
-    topo.layout = "CSR"
     if args.graph_mode == "CUDA" and device.type == "cuda":
-        topo.indptr = indptr
-        topo.indices = col_s
+        topo = Topology((indptr, col_s), input_layout="CSR", layout="CSR",
+                        auto_edge_ids=False)
     else:
-        topo.indptr = indptr.cpu()
-        topo.indices = col_s.cpu()
-    topo.edge_ids = None
-    topo.edge_weights = None
+        topo = Topology((indptr.cpu(), col_s.cpu()), input_layout="CSR",
+                        layout="CSR", auto_edge_ids=False)
     graph = Graph(topo, mode=args.graph_mode, device=device.index)
     if args.graph_mode == "CUDA" and device.type == "cuda":
         # already on device
-        graph._indptr, graph._indices = indptr, col_s
+        graph._indptr, graph._indices = topo.indptr, topo.indices
         graph._edge_ids = graph._edge_weights = None
         graph._lazy_done = True
 
@@ -371,6 +388,7 @@ def main():
                 "feat_dim": args.feat_dim,
                 "classes": args.classes,
                 "graph_mode": args.graph_mode,
+                "graph_skew": args.graph_skew,
                 "parallelism": f"dp{world}",
                 "feature_mode": args.feature_mode,
                 "epoch_time_s_equiv": round(

@@ -46,7 +46,11 @@ class Topology:
     def __init__(self, edge_index, edge_ids: Optional[torch.Tensor] = None,
                  edge_weights: Optional[torch.Tensor] = None,
                  input_layout: str = "COO", layout: str = "CSR",
-                 num_nodes: Optional[int] = None):
+                 num_nodes: Optional[int] = None,
+                 auto_edge_ids: bool = True):
+        """auto_edge_ids=False skips materializing the default arange edge
+        ids (position ids) — saves 8 bytes/edge when the workload never
+        samples with_edge (e.g. bench.py at papers100M scale)."""
         input_layout = input_layout.upper()
         layout = layout.upper()
         assert layout in ("CSR", "CSC")
@@ -69,8 +73,9 @@ class Topology:
         elif input_layout in ("CSR", "CSC"):
             indptr, indices = edge_index
             eids = edge_ids
-            if eids is None:
-                eids = torch.arange(indices.numel(), dtype=torch.long)
+            if eids is None and (auto_edge_ids or input_layout != layout):
+                eids = torch.arange(indices.numel(), dtype=torch.long,
+                                    device=indices.device)
             ew = edge_weights
             if input_layout == layout and not _rows_sorted(indptr, indices):
                 # the negative samplers binary-search within rows
