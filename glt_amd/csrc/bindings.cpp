@@ -211,6 +211,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
 
   m.def("sage_gemm", &hip_sage_gemm, py::arg("A"), py::arg("B"),
         py::arg("bias") = py::none(), py::arg("relu") = false);
+  m.def("gemm_bt_bf16", &hip_gemm_bt_bf16, py::arg("A"), py::arg("Bt"),
+        py::arg("bias") = py::none(), py::arg("relu") = false,
+        py::arg("out_fp32") = false);
+  m.def("gemm_kt_bf16", &hip_gemm_kt_bf16, py::arg("A"), py::arg("B"),
+        py::arg("with_db") = false);
+  m.def("mfma_bf16_selftest", &hip_mfma_bf16_selftest);
   m.def("gat_fused_fwd", &hip_gat_fused_fwd);
   m.def("gat_fused_bwd", &hip_gat_fused_bwd);
   m.def("segment_mean_fwd", &hip_segment_mean_fwd);

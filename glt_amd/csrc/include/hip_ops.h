@@ -117,6 +117,18 @@ torch::Tensor hip_sage_gemm(const torch::Tensor& A, const torch::Tensor& B,
                             const c10::optional<torch::Tensor>& bias,
                             bool relu);
 
+// --- bf16 MFMA projection GEMMs (hip_gemm_bf16.hip) -------------------------
+// C = act(A[M,K] @ Bt[N,K]^T + bias): Bt in nn.Linear weight layout.
+torch::Tensor hip_gemm_bt_bf16(const torch::Tensor& A,
+                               const torch::Tensor& Bt,
+                               const c10::optional<torch::Tensor>& bias,
+                               bool relu, bool out_fp32);
+// (dW, db) = (A[Kb,M]^T @ B[Kb,N], colsum A), fp32 outputs (split-K).
+std::tuple<torch::Tensor, c10::optional<torch::Tensor>> hip_gemm_kt_bf16(
+    const torch::Tensor& A, const torch::Tensor& B, bool with_db);
+torch::Tensor hip_mfma_bf16_selftest(const torch::Tensor& A,
+                                     const torch::Tensor& B);
+
 // --- memory plumbing (hip_mem.hip) -----------------------------------------
 // Device-dtype alias of (pinned/registered) host memory; keeps `src` alive.
 torch::Tensor host_mapped_view(const torch::Tensor& src, int64_t device_index);

@@ -67,10 +67,15 @@ class _CastLinear(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, relu):
         w16 = weight.to(x.dtype)
-        b16 = bias.to(x.dtype) if bias is not None else None
-        out = torch.nn.functional.linear(x, w16, b16)
-        if relu:
-            out = torch.relu_(out)
+        if _use_bf16_mfma(x, weight):
+            from .. import _C
+
+            out = _C.gemm_bt_bf16(x, w16, bias, relu, False)
+        else:
+            b16 = bias.to(x.dtype) if bias is not None else None
+            out = torch.nn.functional.linear(x, w16, b16)
+            if relu:
+                out = torch.relu_(out)
         ctx.save_for_backward(x, w16, out)
         ctx.has_bias = bias is not None
         ctx.relu = relu
@@ -81,15 +86,46 @@ class _CastLinear(torch.autograd.Function):
         x, w16, out = ctx.saved_tensors
         dy = dy.contiguous()
         if ctx.relu:
-            dy = dy * (out > 0).to(dy.dtype)
+            # one-kernel relu mask (vs gt + mul two-kernel form)
+            dy = torch.ops.aten.threshold_backward(dy, out, 0)
         dx = dw = db = None
-        if ctx.needs_input_grad[0]:
-            dx = dy @ w16
-        if ctx.needs_input_grad[1]:
-            dw = (dy.t() @ x).to(torch.float32)
-        if ctx.has_bias and ctx.needs_input_grad[2]:
+        mfma = _use_bf16_mfma(x, w16)
+        if mfma:
+            from .. import _C
+
+            if ctx.needs_input_grad[1]:
+                # fused dW = dy^T @ x and db = colsum(dy), fp32 out
+                dw, db = _C.gemm_kt_bf16(
+                    dy, x, ctx.has_bias and ctx.needs_input_grad[2])
+            if ctx.needs_input_grad[0]:
+                if w16.size(1) % 16 == 0:
+                    dx = _C.gemm_bt_bf16(dy, w16.t().contiguous(), None,
+                                         False, False)
+                else:
+                    dx = dy @ w16
+        else:
+            if ctx.needs_input_grad[0]:
+                dx = dy @ w16
+            if ctx.needs_input_grad[1]:
+                dw = (dy.t() @ x).to(torch.float32)
+            if ctx.has_bias and ctx.needs_input_grad[2]:
+                db = dy.sum(0).to(torch.float32)
+        if ctx.has_bias and ctx.needs_input_grad[2] and db is None:
             db = dy.sum(0).to(torch.float32)
         return dx, dw, db, None
+
+
+def _use_bf16_mfma(x: torch.Tensor, weight: torch.Tensor) -> bool:
+    """Dispatch policy for the hand-written bf16 MFMA GEMMs.  Requires
+    device bf16; small-K/N tails are handled by the kernels, but tiny
+    classifier heads (N<64) stay on hipBLASLt where the custom tiling
+    has no parallelism to win with."""
+    import os
+
+    if os.environ.get("GLT_DISABLE_BF16_MFMA"):
+        return False
+    return (x.is_cuda and x.dtype == torch.bfloat16
+            and weight.size(0) >= 32)
 
 
 def cast_linear(x: torch.Tensor, weight: torch.Tensor,

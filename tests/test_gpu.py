@@ -758,3 +758,89 @@ def test_end_to_end_learning_sbm_bf16():
                 total += y.numel()
     acc = correct / max(total, 1)
     assert acc > 0.85, f"bf16 end-to-end accuracy too low: {acc}"
+
+
+def test_mfma_bf16_selftest_layout():
+    """Ground-truth fragment layout check for mfma_f32_16x16x32_bf16
+    (asymmetric A and B per the guide's A=I-check rule)."""
+    A = (torch.arange(16 * 32, device="cuda").float()
+         .reshape(16, 32) % 7 - 3).to(torch.bfloat16)
+    B = (torch.arange(32 * 16, device="cuda").float()
+         .reshape(32, 16) % 5 - 2).to(torch.bfloat16)
+    C = _C.mfma_bf16_selftest(A, B)
+    ref = A.float() @ B.float()
+    assert torch.allclose(C, ref, atol=1e-3), (C - ref).abs().max()
+
+
+def test_gemm_bt_bf16_numerics():
+    """C = relu(A @ W^T + bias) vs fp32 torch on the flagship shapes
+    incl. K/N/M tails."""
+    torch.manual_seed(0)
+    for M, K, N in [(1000, 200, 256), (257, 512, 256), (64, 100, 47),
+                    (130, 33, 64), (4, 16, 64)]:
+        A = torch.randn(M, K, device="cuda").to(torch.bfloat16)
+        W = (torch.randn(N, K, device="cuda") / K ** 0.5).to(
+            torch.bfloat16)
+        b = torch.randn(N, device="cuda")
+        C = _C.gemm_bt_bf16(A, W, b, True, False)
+        ref = torch.relu(A.float() @ W.float().t() + b)
+        scale = max(ref.abs().max().item(), 1.0)
+        assert (C.float() - ref).abs().max() / scale < 2e-2, \
+            (M, K, N, (C.float() - ref).abs().max().item())
+        # fp32-out form
+        C32 = _C.gemm_bt_bf16(A, W, None, False, True)
+        ref32 = A.float() @ W.float().t()
+        assert C32.dtype == torch.float32
+        assert (C32 - ref32).abs().max() / scale < 2e-2
+
+
+def test_gemm_kt_bf16_numerics():
+    """dW = A^T @ B with fused db colsum vs fp32 torch (split-K path:
+    long-K accumulation must stay fp32-exact in ordering tolerance)."""
+    torch.manual_seed(1)
+    for Kb, M, N in [(20_000, 256, 200), (1000, 47, 512), (64, 64, 64),
+                     (130, 40, 30)]:
+        A = torch.randn(Kb, M, device="cuda").to(torch.bfloat16)
+        B = torch.randn(Kb, N, device="cuda").to(torch.bfloat16)
+        C, db = _C.gemm_kt_bf16(A, B, True)
+        ref = A.float().t() @ B.float()
+        dbr = A.float().sum(0)
+        rel = (C - ref).abs().max().item() / max(ref.abs().max().item(),
+                                                 1.0)
+        assert rel < 2e-3, (Kb, M, N, rel)
+        dbrel = (db - dbr).abs().max().item() / max(
+            dbr.abs().max().item(), 1.0)
+        assert dbrel < 1e-3, (Kb, M, N, dbrel)
+        C2, none = _C.gemm_kt_bf16(A, B, False)
+        assert none is None and torch.equal(C2, C) or \
+            (C2 - C).abs().max() < 1e-3
+
+
+def test_cast_linear_gpu_mfma_grads():
+    """cast_linear on GPU routes through the MFMA kernels; grads match
+    the plain fp32 reference at bf16 tolerance."""
+    from glt_amd.ops import cast_linear
+
+    torch.manual_seed(2)
+    M, K, N = 5000, 200, 256
+    w = torch.randn(N, K, device="cuda", requires_grad=True)
+    w.data /= K ** 0.5
+    b = torch.zeros(N, device="cuda", requires_grad=True)
+    x = torch.randn(M, K, device="cuda").to(torch.bfloat16
+                                            ).requires_grad_(True)
+    out = cast_linear(x, w, b, relu=True)
+    g = torch.randn_like(out)
+    out.backward(g)
+
+    w2 = w.detach().clone().requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+    x2 = x.detach().float().requires_grad_(True)
+    ref = torch.relu(torch.nn.functional.linear(x2, w2, b2))
+    ref.backward(g.float())
+    assert (out.float() - ref).abs().max() < 0.1
+    wscale = max(w2.grad.abs().max().item(), 1.0)
+    assert (w.grad - w2.grad).abs().max() / wscale < 2e-2
+    assert (b.grad - b2.grad).abs().max() / max(
+        b2.grad.abs().max().item(), 1.0) < 2e-2
+    assert (x.grad.float() - x2.grad).abs().max() / max(
+        x2.grad.abs().max().item(), 1.0) < 5e-2
