@@ -15,6 +15,8 @@
  * lose too much precision for degree-weighted sums); the python wrapper
  * casts once at the end.
  */
+#include <hip/hip_bf16.h>
+
 #include "hip_common.h"
 #include "../include/common.h"
 
@@ -42,6 +44,60 @@ __global__ void seg_mean_fwd_kernel(const scalar_t* __restrict__ x,
       for (int64_t k = s; k < e; ++k)
         acc += static_cast<float>(x[col[k] * feat + f]);
       out[t * feat + f] = static_cast<scalar_t>(acc * inv);
+    }
+  }
+}
+
+// bf16 fast path: each lane owns TWO adjacent channels (one 4-byte
+// ushort2 load per row visit) — the scalar template's 2-byte gathers
+// were measured at only 1.36x fp32 on the HBM-bound forward.
+__global__ void seg_mean_fwd_bf162_kernel(const __hip_bfloat162* __restrict__ x,
+                                          const int64_t* __restrict__ col,
+                                          const int64_t* __restrict__ off,
+                                          int64_t n_tgt, int64_t feat2,
+                                          __hip_bfloat162* __restrict__ out) {
+  const int lane = threadIdx.x & (kWave - 1);
+  const int64_t wave =
+      (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) / kWave;
+  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) / kWave;
+  for (int64_t t = wave; t < n_tgt; t += n_waves) {
+    const int64_t s = off[t], e = off[t + 1];
+    const float inv = e > s ? 1.0f / (float)(e - s) : 0.0f;
+    for (int64_t f = lane; f < feat2; f += kWave) {
+      float2 acc = {0.f, 0.f};
+      for (int64_t k = s; k < e; ++k) {
+        const float2 v = __bfloat1622float2(x[col[k] * feat2 + f]);
+        acc.x += v.x;
+        acc.y += v.y;
+      }
+      out[t * feat2 + f] =
+          __float22bfloat162_rn({acc.x * inv, acc.y * inv});
+    }
+  }
+}
+
+__global__ void seg_mean_cat_fwd_bf162_kernel(
+    const __hip_bfloat162* __restrict__ x,
+    const int64_t* __restrict__ col, const int64_t* __restrict__ off,
+    int64_t n_tgt, int64_t feat2, __hip_bfloat162* __restrict__ out) {
+  const int lane = threadIdx.x & (kWave - 1);
+  const int64_t wave =
+      (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) / kWave;
+  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) / kWave;
+  const int64_t ostride = 2 * feat2;
+  for (int64_t t = wave; t < n_tgt; t += n_waves) {
+    const int64_t s = off[t], e = off[t + 1];
+    const float inv = e > s ? 1.0f / (float)(e - s) : 0.0f;
+    for (int64_t f = lane; f < feat2; f += kWave) {
+      float2 acc = {0.f, 0.f};
+      for (int64_t k = s; k < e; ++k) {
+        const float2 v = __bfloat1622float2(x[col[k] * feat2 + f]);
+        acc.x += v.x;
+        acc.y += v.y;
+      }
+      out[t * ostride + f] =
+          __float22bfloat162_rn({acc.x * inv, acc.y * inv});
+      out[t * ostride + feat2 + f] = x[t * feat2 + f];
     }
   }
 }
@@ -161,6 +217,18 @@ torch::Tensor hip_segment_mean_fwd(const torch::Tensor& x,
   const int64_t feat = x.size(1);
   auto out = torch::empty({n_tgt, feat}, x.options());
   if (n_tgt > 0) {
+    if (x.scalar_type() == torch::kBFloat16 && feat % 2 == 0) {
+      hipLaunchKernelGGL(seg_mean_fwd_bf162_kernel,
+                         dim3(wave_grid(n_tgt)), dim3(kBlock), 0,
+                         current_stream(),
+                         reinterpret_cast<const __hip_bfloat162*>(
+                             x.data_ptr()),
+                         col.data_ptr<int64_t>(),
+                         offsets.data_ptr<int64_t>(), n_tgt, feat / 2,
+                         reinterpret_cast<__hip_bfloat162*>(
+                             out.data_ptr()));
+      return out;
+    }
     GLT_DISPATCH_SEG(x.scalar_type(), "segment_mean_fwd", [&] {
       hipLaunchKernelGGL(seg_mean_fwd_kernel<scalar_t>,
                          dim3(wave_grid(n_tgt)), dim3(kBlock), 0,
@@ -184,6 +252,18 @@ torch::Tensor hip_segment_mean_cat_fwd(const torch::Tensor& x,
   const int64_t feat = x.size(1);
   auto out = torch::empty({n_tgt, 2 * feat}, x.options());
   if (n_tgt > 0) {
+    if (x.scalar_type() == torch::kBFloat16 && feat % 2 == 0) {
+      hipLaunchKernelGGL(seg_mean_cat_fwd_bf162_kernel,
+                         dim3(wave_grid(n_tgt)), dim3(kBlock), 0,
+                         current_stream(),
+                         reinterpret_cast<const __hip_bfloat162*>(
+                             x.data_ptr()),
+                         col.data_ptr<int64_t>(),
+                         offsets.data_ptr<int64_t>(), n_tgt, feat / 2,
+                         reinterpret_cast<__hip_bfloat162*>(
+                             out.data_ptr()));
+      return out;
+    }
     GLT_DISPATCH_SEG(x.scalar_type(), "segment_mean_cat_fwd", [&] {
       hipLaunchKernelGGL(seg_mean_cat_fwd_kernel<scalar_t>,
                          dim3(wave_grid(n_tgt)), dim3(kBlock), 0,
