@@ -12,9 +12,11 @@
  *       epilogue (kills the separate activation round-trip).
  *  2. gemm_kt_kernel:  dW[M,N] = A[Kb,M]^T @ B[Kb,N], db[M] = colsum(A)
  *     - the weight-gradient shape (K huge = batch rows): split-K over
- *       grid.z with fp32 atomics into the output; the bias gradient
- *       falls out of the same dy tiles for free (the separate
- *       [rows,256] bf16 column-reduce measured 78 us/call).
+ *       grid.z into per-chunk partial planes (a torch sum folds them);
+ *       the bias gradient rides the same kernel (the separate
+ *       [rows,256] bf16 column-reduce measured 78 us/call).  Runs on
+ *       the f32 16x16x4 MFMA whose k-per-lane-group operand layout
+ *       matches the [k][cols] inputs with no LDS transpose.
  *     - fp32 out: master-grad dtype, so the .to(fp32) casts disappear.
  *  3. mfma_bf16_selftest: one 16x16x32 / 32x32x16 tile from explicit
  *     matrices (fragment-layout ground truth for the GPU tests).
@@ -171,8 +173,15 @@ void gemm_bt_kernel(const __bf16* __restrict__ A,
 
 // ---------------------------------------------------------------------------
 // dW[M,N] = A[Kb,M]^T @ B[Kb,N] (+ db[M] = colsum A), split-K over
-// grid.z, fp32 atomicAdd outputs.  A/B tiles are transposed during the
-// LDS write pass so fragments read as contiguous 16-byte vectors.
+// grid.z with per-chunk partial planes.
+//
+// Uses the f32 16x16x4 MFMA on purpose: its operand layout is
+// k-per-lane-GROUP (lane l: A[i=l&15][k=l>>4]), which matches the
+// [k][cols] memory layout of both inputs EXACTLY -- fragments read from
+// a row-major LDS image with zero transpose work (the bf16 16x16x32
+// form needs k-contiguous-per-lane fragments, i.e. a full LDS
+// transpose: measured 56 TF/s staging-bound).  bf16->f32 conversion
+// happens once per element in the staging write pass.
 // ---------------------------------------------------------------------------
 template <bool WITH_DB>
 __global__ __launch_bounds__(256)
@@ -181,8 +190,9 @@ void gemm_kt_kernel(const __bf16* __restrict__ A,
                     float* __restrict__ C,
                     float* __restrict__ db,
                     int64_t Kb, int64_t M, int64_t N, int64_t k_per_z) {
-  __shared__ __bf16 As[GBM][GBK + LDP];   // As[m][k]
-  __shared__ __bf16 Bs[GBN][GBK + LDP];   // Bs[n][k]
+  constexpr int KT = 32;                 // k-depth per LDS stage
+  __shared__ float As[KT][GBM + 4];      // As[k][m], f32
+  __shared__ float Bs[KT][GBN + 4];      // Bs[k][n], f32
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -197,80 +207,62 @@ void gemm_kt_kernel(const __bf16* __restrict__ A,
 
   f32x4b acc[2][2] = {};
 
-  // staging map (transposing): 256 threads; thread handles global row
-  // kz+t_k, 16 consecutive m (or n) columns; writes 16 strided LDS b16.
-  const int t_k = tid >> 2;            // 0..63 (k within tile)
-  const int t_c = (tid & 3) * 16;      // column base 0,16,32,48
+  // staging: 256 threads cover a [KT=32][64] tile as 8 elements each
+  // (one bf16x8 global load, 8 f32 LDS writes).
+  const int t_k = tid >> 3;            // 0..31
+  const int t_c = (tid & 7) * 8;       // 0..56 step 8
 
-  const int fi = lane & 15;
-  const int fk8 = (lane >> 4) * 8;
+  const int fi = lane & 15;            // fragment row/col
+  const int fk = lane >> 4;            // fragment k sub-index (0..3)
 
-  for (int64_t k0 = kz0; k0 < kz1; k0 += GBK) {
+  for (int64_t k0 = kz0; k0 < kz1; k0 += KT) {
     const int64_t g_k = k0 + t_k;
-    // stage A^T: As[m][k] = A[k][m]
     {
-      bf16x8 v0 = {}, v1 = {};
+      bf16x8 v = {};
       if (g_k < kz1 && block_row + t_c < M) {
         const int64_t base = g_k * M + block_row + t_c;
-        if (block_row + t_c + 15 < M) {
-          v0 = *reinterpret_cast<const bf16x8*>(&A[base]);
-          v1 = *reinterpret_cast<const bf16x8*>(&A[base + 8]);
+        if (block_row + t_c + 7 < M) {
+          v = *reinterpret_cast<const bf16x8*>(&A[base]);
         } else {
 #pragma unroll
-          for (int e = 0; e < 8; ++e) {
-            if (block_row + t_c + e < M) v0[e] = A[base + e];
-            if (block_row + t_c + 8 + e < M) v1[e] = A[base + 8 + e];
-          }
+          for (int e = 0; e < 8; ++e)
+            if (block_row + t_c + e < M) v[e] = A[base + e];
         }
       }
 #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        As[t_c + e][t_k] = v0[e];
-        As[t_c + 8 + e][t_k] = v1[e];
-      }
+      for (int e = 0; e < 8; ++e) As[t_k][t_c + e] = bf2f(v[e]);
     }
-    // stage B^T: Bs[n][k] = B[k][n]
     {
-      bf16x8 v0 = {}, v1 = {};
+      bf16x8 v = {};
       if (g_k < kz1 && block_col + t_c < N) {
         const int64_t base = g_k * N + block_col + t_c;
-        if (block_col + t_c + 15 < N) {
-          v0 = *reinterpret_cast<const bf16x8*>(&B[base]);
-          v1 = *reinterpret_cast<const bf16x8*>(&B[base + 8]);
+        if (block_col + t_c + 7 < N) {
+          v = *reinterpret_cast<const bf16x8*>(&B[base]);
         } else {
 #pragma unroll
-          for (int e = 0; e < 8; ++e) {
-            if (block_col + t_c + e < N) v0[e] = B[base + e];
-            if (block_col + t_c + 8 + e < N) v1[e] = B[base + 8 + e];
-          }
+          for (int e = 0; e < 8; ++e)
+            if (block_col + t_c + e < N) v[e] = B[base + e];
         }
       }
 #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        Bs[t_c + e][t_k] = v0[e];
-        Bs[t_c + 8 + e][t_k] = v1[e];
-      }
+      for (int e = 0; e < 8; ++e) Bs[t_k][t_c + e] = bf2f(v[e]);
     }
     __syncthreads();
 
 #pragma unroll
-    for (int kk = 0; kk < GBK; kk += 32) {
-      bf16x8 a0 = *reinterpret_cast<const bf16x8*>(
-          &As[wr * 32 + fi][kk + fk8]);
-      bf16x8 a1 = *reinterpret_cast<const bf16x8*>(
-          &As[wr * 32 + 16 + fi][kk + fk8]);
-      bf16x8 b0 = *reinterpret_cast<const bf16x8*>(
-          &Bs[wc * 32 + fi][kk + fk8]);
-      bf16x8 b1 = *reinterpret_cast<const bf16x8*>(
-          &Bs[wc * 32 + 16 + fi][kk + fk8]);
-      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0,
-                                                          acc[0][0], 0, 0, 0);
-      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1,
-                                                          acc[0][1], 0, 0, 0);
-      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0,
-                                                          acc[1][0], 0, 0, 0);
-      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1,
-                                                          acc[1][1], 0, 0, 0);
+    for (int kk = 0; kk < KT; kk += 4) {
+      const float a0 = As[kk + fk][wr * 32 + fi];
+      const float a1 = As[kk + fk][wr * 32 + 16 + fi];
+      const float b0 = Bs[kk + fk][wc * 32 + fi];
+      const float b1 = Bs[kk + fk][wc * 32 + 16 + fi];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc[0][0],
+                                                       0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc[0][1],
+                                                       0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc[1][0],
+                                                       0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc[1][1],
+                                                       0, 0, 0);
     }
     __syncthreads();
   }

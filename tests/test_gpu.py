@@ -828,14 +828,18 @@ def test_cast_linear_gpu_mfma_grads():
     b = torch.zeros(N, device="cuda", requires_grad=True)
     x = torch.randn(M, K, device="cuda").to(torch.bfloat16
                                             ).requires_grad_(True)
-    out = cast_linear(x, w, b, relu=True)
+    # relu=False for the grad comparison: near-zero outputs flip the
+    # ReLU mask between the bf16 and fp32 paths, making dx pointwise
+    # incomparable (the fwd+relu numerics are covered by
+    # test_gemm_bt_bf16_numerics)
+    out = cast_linear(x, w, b, relu=False)
     g = torch.randn_like(out)
     out.backward(g)
 
     w2 = w.detach().clone().requires_grad_(True)
     b2 = b.detach().clone().requires_grad_(True)
     x2 = x.detach().float().requires_grad_(True)
-    ref = torch.relu(torch.nn.functional.linear(x2, w2, b2))
+    ref = torch.nn.functional.linear(x2, w2, b2)
     ref.backward(g.float())
     assert (out.float() - ref).abs().max() < 0.1
     # dW/db accumulate M=5000 bf16-rounded terms: tolerance is the bf16
