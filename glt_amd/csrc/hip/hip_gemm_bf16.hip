@@ -173,15 +173,11 @@ void gemm_bt_kernel(const __bf16* __restrict__ A,
 
 // ---------------------------------------------------------------------------
 // dW[M,N] = A[Kb,M]^T @ B[Kb,N] (+ db[M] = colsum A), split-K over
-// grid.z with per-chunk partial planes.
-//
-// Uses the f32 16x16x4 MFMA on purpose: its operand layout is
-// k-per-lane-GROUP (lane l: A[i=l&15][k=l>>4]), which matches the
-// [k][cols] memory layout of both inputs EXACTLY -- fragments read from
-// a row-major LDS image with zero transpose work (the bf16 16x16x32
-// form needs k-contiguous-per-lane fragments, i.e. a full LDS
-// transpose: measured 56 TF/s staging-bound).  bf16->f32 conversion
-// happens once per element in the staging write pass.
+// grid.z with per-chunk partial planes.  A/B tiles are transposed
+// during the LDS write pass (paired k-rows -> b32 writes) so fragments
+// read as contiguous 16-byte vectors for the bf16 16x16x32 MFMA.
+// (An f32-16x16x4 variant with transpose-free fragments measured
+// SLOWER — 40 vs 56 TF/s — the f32 MFMA issue rate is the wall.)
 // ---------------------------------------------------------------------------
 template <bool WITH_DB>
 __global__ __launch_bounds__(256)
@@ -190,9 +186,8 @@ void gemm_kt_kernel(const __bf16* __restrict__ A,
                     float* __restrict__ C,
                     float* __restrict__ db,
                     int64_t Kb, int64_t M, int64_t N, int64_t k_per_z) {
-  constexpr int KT = 32;                 // k-depth per LDS stage
-  __shared__ float As[KT][GBM + 4];      // As[k][m], f32
-  __shared__ float Bs[KT][GBN + 4];      // Bs[k][n], f32
+  __shared__ __bf16 As[GBM][GBK + LDP];   // As[m][k]
+  __shared__ __bf16 Bs[GBN][GBK + LDP];   // Bs[n][k]
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -207,62 +202,89 @@ void gemm_kt_kernel(const __bf16* __restrict__ A,
 
   f32x4b acc[2][2] = {};
 
-  // staging: 256 threads cover a [KT=32][64] tile as 8 elements each
-  // (one bf16x8 global load, 8 f32 LDS writes).
-  const int t_k = tid >> 3;            // 0..31
+  // transposing stage: thread owns TWO adjacent k rows x 8 cols, so
+  // each LDS write is a b32 pair (halves the transpose instruction
+  // count vs scalar b16 writes).  32 k-pairs x 8 col-groups = 256.
+  const int t_k = (tid >> 3) * 2;      // 0..62 step 2
   const int t_c = (tid & 7) * 8;       // 0..56 step 8
 
-  const int fi = lane & 15;            // fragment row/col
-  const int fk = lane >> 4;            // fragment k sub-index (0..3)
+  const int fi = lane & 15;
+  const int fk8 = (lane >> 4) * 8;
 
-  for (int64_t k0 = kz0; k0 < kz1; k0 += KT) {
+  for (int64_t k0 = kz0; k0 < kz1; k0 += GBK) {
     const int64_t g_k = k0 + t_k;
+    // stage A^T: As[m][k] = A[k][m]
     {
-      bf16x8 v = {};
+      bf16x8 v0 = {}, v1 = {};
       if (g_k < kz1 && block_row + t_c < M) {
         const int64_t base = g_k * M + block_row + t_c;
         if (block_row + t_c + 7 < M) {
-          v = *reinterpret_cast<const bf16x8*>(&A[base]);
+          v0 = *reinterpret_cast<const bf16x8*>(&A[base]);
+          if (g_k + 1 < kz1)
+            v1 = *reinterpret_cast<const bf16x8*>(&A[base + M]);
         } else {
 #pragma unroll
-          for (int e = 0; e < 8; ++e)
-            if (block_row + t_c + e < M) v[e] = A[base + e];
+          for (int e = 0; e < 8; ++e) {
+            if (block_row + t_c + e < M) {
+              v0[e] = A[base + e];
+              if (g_k + 1 < kz1) v1[e] = A[base + M + e];
+            }
+          }
         }
       }
 #pragma unroll
-      for (int e = 0; e < 8; ++e) As[t_k][t_c + e] = bf2f(v[e]);
+      for (int e = 0; e < 8; ++e) {
+        __bf16 pair[2] = {v0[e], v1[e]};
+        *reinterpret_cast<uint32_t*>(&As[t_c + e][t_k]) =
+            *reinterpret_cast<const uint32_t*>(pair);
+      }
     }
+    // stage B^T: Bs[n][k] = B[k][n]
     {
-      bf16x8 v = {};
+      bf16x8 v0 = {}, v1 = {};
       if (g_k < kz1 && block_col + t_c < N) {
         const int64_t base = g_k * N + block_col + t_c;
         if (block_col + t_c + 7 < N) {
-          v = *reinterpret_cast<const bf16x8*>(&B[base]);
+          v0 = *reinterpret_cast<const bf16x8*>(&B[base]);
+          if (g_k + 1 < kz1)
+            v1 = *reinterpret_cast<const bf16x8*>(&B[base + N]);
         } else {
 #pragma unroll
-          for (int e = 0; e < 8; ++e)
-            if (block_col + t_c + e < N) v[e] = B[base + e];
+          for (int e = 0; e < 8; ++e) {
+            if (block_col + t_c + e < N) {
+              v0[e] = B[base + e];
+              if (g_k + 1 < kz1) v1[e] = B[base + N + e];
+            }
+          }
         }
       }
 #pragma unroll
-      for (int e = 0; e < 8; ++e) Bs[t_k][t_c + e] = bf2f(v[e]);
+      for (int e = 0; e < 8; ++e) {
+        __bf16 pair[2] = {v0[e], v1[e]};
+        *reinterpret_cast<uint32_t*>(&Bs[t_c + e][t_k]) =
+            *reinterpret_cast<const uint32_t*>(pair);
+      }
     }
     __syncthreads();
 
 #pragma unroll
-    for (int kk = 0; kk < KT; kk += 4) {
-      const float a0 = As[kk + fk][wr * 32 + fi];
-      const float a1 = As[kk + fk][wr * 32 + 16 + fi];
-      const float b0 = Bs[kk + fk][wc * 32 + fi];
-      const float b1 = Bs[kk + fk][wc * 32 + 16 + fi];
-      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc[0][0],
-                                                       0, 0, 0);
-      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc[0][1],
-                                                       0, 0, 0);
-      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc[1][0],
-                                                       0, 0, 0);
-      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc[1][1],
-                                                       0, 0, 0);
+    for (int kk = 0; kk < GBK; kk += 32) {
+      bf16x8 a0 = *reinterpret_cast<const bf16x8*>(
+          &As[wr * 32 + fi][kk + fk8]);
+      bf16x8 a1 = *reinterpret_cast<const bf16x8*>(
+          &As[wr * 32 + 16 + fi][kk + fk8]);
+      bf16x8 b0 = *reinterpret_cast<const bf16x8*>(
+          &Bs[wc * 32 + fi][kk + fk8]);
+      bf16x8 b1 = *reinterpret_cast<const bf16x8*>(
+          &Bs[wc * 32 + 16 + fi][kk + fk8]);
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0,
+                                                          acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1,
+                                                          acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0,
+                                                          acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1,
+                                                          acc[1][1], 0, 0, 0);
     }
     __syncthreads();
   }
