@@ -327,6 +327,150 @@ void gemm_kt_kernel(const __bf16* __restrict__ A,
 }
 
 // ---------------------------------------------------------------------------
+// 128x128 / 8-wave variant of gemm_kt: halves the A/B re-read
+// amplification (traffic ∝ tiles-in-other-dim) for the big dW shapes.
+// Waves as 2(m)x4(n), each wave 64x32 = 4x2 16x16 fragments.
+// ---------------------------------------------------------------------------
+constexpr int KBM = 128, KBN = 128;
+
+template <bool WITH_DB>
+__global__ __launch_bounds__(512)
+void gemm_kt128_kernel(const __bf16* __restrict__ A,
+                       const __bf16* __restrict__ B,
+                       float* __restrict__ C,
+                       float* __restrict__ db,
+                       int64_t Kb, int64_t M, int64_t N,
+                       int64_t k_per_z) {
+  __shared__ __bf16 As[KBM][GBK + LDP];   // As[m][k]
+  __shared__ __bf16 Bs[KBN][GBK + LDP];   // Bs[n][k]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;               // 0..7
+  const int wm = wave >> 2;                // 0..1 (64 rows each)
+  const int wn = wave & 3;                 // 0..3 (32 cols each)
+
+  const int64_t block_row = (int64_t)blockIdx.x * KBM;
+  const int64_t block_col = (int64_t)blockIdx.y * KBN;
+  const int64_t kz0 = (int64_t)blockIdx.z * k_per_z;
+  const int64_t kz1 = std::min(kz0 + k_per_z, Kb);
+
+  f32x4b acc[4][2] = {};
+
+  // transposing stage: 512 threads = 16 col-groups x 32 k-pairs
+  const int t_k = (tid >> 4) * 2;          // 0..62 step 2
+  const int t_c = (tid & 15) * 8;          // 0..120 step 8
+
+  const int fi = lane & 15;
+  const int fk8 = (lane >> 4) * 8;
+
+  for (int64_t k0 = kz0; k0 < kz1; k0 += GBK) {
+    const int64_t g_k = k0 + t_k;
+    {
+      bf16x8 v0 = {}, v1 = {};
+      if (g_k < kz1 && block_row + t_c < M) {
+        const int64_t base = g_k * M + block_row + t_c;
+        if (block_row + t_c + 7 < M) {
+          v0 = *reinterpret_cast<const bf16x8*>(&A[base]);
+          if (g_k + 1 < kz1)
+            v1 = *reinterpret_cast<const bf16x8*>(&A[base + M]);
+        } else {
+#pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            if (block_row + t_c + e < M) {
+              v0[e] = A[base + e];
+              if (g_k + 1 < kz1) v1[e] = A[base + M + e];
+            }
+          }
+        }
+      }
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        __bf16 pair[2] = {v0[e], v1[e]};
+        *reinterpret_cast<uint32_t*>(&As[t_c + e][t_k]) =
+            *reinterpret_cast<const uint32_t*>(pair);
+      }
+    }
+    {
+      bf16x8 v0 = {}, v1 = {};
+      if (g_k < kz1 && block_col + t_c < N) {
+        const int64_t base = g_k * N + block_col + t_c;
+        if (block_col + t_c + 7 < N) {
+          v0 = *reinterpret_cast<const bf16x8*>(&B[base]);
+          if (g_k + 1 < kz1)
+            v1 = *reinterpret_cast<const bf16x8*>(&B[base + N]);
+        } else {
+#pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            if (block_col + t_c + e < N) {
+              v0[e] = B[base + e];
+              if (g_k + 1 < kz1) v1[e] = B[base + N + e];
+            }
+          }
+        }
+      }
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        __bf16 pair[2] = {v0[e], v1[e]};
+        *reinterpret_cast<uint32_t*>(&Bs[t_c + e][t_k]) =
+            *reinterpret_cast<const uint32_t*>(pair);
+      }
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int kk = 0; kk < GBK; kk += 32) {
+      bf16x8 a[4], b[2];
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+        a[i] = *reinterpret_cast<const bf16x8*>(
+            &As[wm * 64 + i * 16 + fi][kk + fk8]);
+#pragma unroll
+      for (int j = 0; j < 2; ++j)
+        b[j] = *reinterpret_cast<const bf16x8*>(
+            &Bs[wn * 32 + j * 16 + fi][kk + fk8]);
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int c_col = lane & 15;
+  const int c_row0 = (lane >> 4) * 4;
+  float* Cz = C + (int64_t)blockIdx.z * M * N;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int64_t row = block_row + wm * 64 + i * 16 + c_row0 + r;
+        const int64_t col = block_col + wn * 32 + j * 16 + c_col;
+        if (row < M && col < N) Cz[row * N + col] = acc[i][j][r];
+      }
+    }
+  }
+  if (WITH_DB && blockIdx.y == 0) {
+    __shared__ float dbs[KBM][4];
+    const int m_l = tid & 127, phase = tid >> 7;
+    float s = 0.f;
+    if (block_row + m_l < M) {
+      for (int64_t k = kz0 + phase; k < kz1; k += 4)
+        s += bf2f(A[k * M + block_row + m_l]);
+    }
+    dbs[m_l][phase] = s;
+    __syncthreads();
+    if (tid < KBM && block_row + tid < M)
+      atomicAdd(&db[block_row + tid], dbs[tid][0] + dbs[tid][1] +
+                                          dbs[tid][2] + dbs[tid][3]);
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Single-tile fragment-layout selftests.
 // ---------------------------------------------------------------------------
 __global__ void selftest_16x16x32(const __bf16* A, const __bf16* B,
@@ -399,11 +543,16 @@ std::tuple<torch::Tensor, c10::optional<torch::Tensor>> hip_gemm_kt_bf16(
   const int64_t Kb = A.size(0), M = A.size(1), N = B.size(1);
   auto Ac = A.contiguous();
   auto Bc = B.contiguous();
-  const int64_t m_t = (M + GBM - 1) / GBM, n_t = (N + GBN - 1) / GBN;
+  // 128x128/8-wave tiles for big dW shapes (halves the A/B re-read
+  // amplification); 64x64 for small/tail-heavy outputs
+  const bool big = (M >= KBM && N >= 96 && Kb >= 4096);
+  const int64_t bm = big ? KBM : GBM, bn = big ? KBN : GBN;
+  const int64_t m_t = (M + bm - 1) / bm, n_t = (N + bn - 1) / bn;
   // split-K sized so the grid fills the 256 CUs a few times over
   int64_t z = 1;
   if (Kb > GBK) {
-    const int64_t want = (1024 + m_t * n_t - 1) / (m_t * n_t);
+    const int64_t want_wg = big ? 512 : 1024;
+    const int64_t want = (want_wg + m_t * n_t - 1) / (m_t * n_t);
     const int64_t max_z = (Kb + GBK - 1) / GBK;
     z = std::max<int64_t>(1, std::min(want, max_z));
   }
@@ -421,11 +570,20 @@ std::tuple<torch::Tensor, c10::optional<torch::Tensor>> hip_gemm_kt_bf16(
   // z>1: per-chunk partial planes + one sum(0) (no output atomics)
   auto P = torch::empty({z, M, N}, A.options().dtype(torch::kFloat32));
   dim3 grid((uint32_t)m_t, (uint32_t)n_t, (uint32_t)z);
-  auto* kfn = with_db ? gemm_kt_kernel<true> : gemm_kt_kernel<false>;
-  hipLaunchKernelGGL(kfn, grid, dim3(256), 0, current_stream(),
-                     reinterpret_cast<const __bf16*>(Ac.data_ptr()),
-                     reinterpret_cast<const __bf16*>(Bc.data_ptr()),
-                     P.data_ptr<float>(), db_p, Kb, M, N, k_per_z);
+  if (big) {
+    auto* kfn = with_db ? gemm_kt128_kernel<true>
+                        : gemm_kt128_kernel<false>;
+    hipLaunchKernelGGL(kfn, grid, dim3(512), 0, current_stream(),
+                       reinterpret_cast<const __bf16*>(Ac.data_ptr()),
+                       reinterpret_cast<const __bf16*>(Bc.data_ptr()),
+                       P.data_ptr<float>(), db_p, Kb, M, N, k_per_z);
+  } else {
+    auto* kfn = with_db ? gemm_kt_kernel<true> : gemm_kt_kernel<false>;
+    hipLaunchKernelGGL(kfn, grid, dim3(256), 0, current_stream(),
+                       reinterpret_cast<const __bf16*>(Ac.data_ptr()),
+                       reinterpret_cast<const __bf16*>(Bc.data_ptr()),
+                       P.data_ptr<float>(), db_p, Kb, M, N, k_per_z);
+  }
   auto C = z == 1 ? P.squeeze(0) : P.sum(0);
   return {C, db};
 }
