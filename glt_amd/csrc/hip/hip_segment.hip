@@ -184,64 +184,6 @@ __global__ void seg_mean_cat_bwd_kernel(const scalar_t* __restrict__ dy,
   }
 }
 
-// bf16 fast-path backward twins: dy loads as __hip_bfloat162 (4-byte),
-// atomics stay fp32 per channel (the arena is fp32).
-__global__ void seg_mean_bwd_bf162_kernel(
-    const __hip_bfloat162* __restrict__ dy,
-    const int64_t* __restrict__ col, const int64_t* __restrict__ off,
-    int64_t n_tgt, int64_t feat2, float* __restrict__ dx) {
-  const int lane = threadIdx.x & (kWave - 1);
-  const int64_t wave =
-      (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) / kWave;
-  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) / kWave;
-  for (int64_t t = wave; t < n_tgt; t += n_waves) {
-    const int64_t s = off[t], e = off[t + 1];
-    if (e <= s) continue;
-    const float inv = 1.0f / (float)(e - s);
-    for (int64_t f = lane; f < feat2; f += kWave) {
-      const float2 g = __bfloat1622float2(dy[t * feat2 + f]);
-      for (int64_t k = s; k < e; ++k) {
-        const int64_t c = col[k];
-        atomicAdd(&dx[c * 2 * feat2 + 2 * f], g.x * inv);
-        atomicAdd(&dx[c * 2 * feat2 + 2 * f + 1], g.y * inv);
-      }
-    }
-  }
-}
-
-__global__ void seg_mean_cat_bwd_bf162_kernel(
-    const __hip_bfloat162* __restrict__ dy,
-    const int64_t* __restrict__ col, const int64_t* __restrict__ off,
-    int64_t n_tgt, int64_t feat2, float* __restrict__ dx) {
-  const int lane = threadIdx.x & (kWave - 1);
-  const int64_t wave =
-      (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) / kWave;
-  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) / kWave;
-  const int64_t ostride = 2 * feat2;
-  for (int64_t t = wave; t < n_tgt; t += n_waves) {
-    const int64_t s = off[t], e = off[t + 1];
-    for (int64_t f = lane; f < feat2; f += kWave) {
-      const float2 gr = __bfloat1622float2(dy[t * ostride + feat2 + f]);
-      // root rows are unique per t in the sorted-batch convention, but
-      // dx may also receive neighbor contributions: atomics keep it
-      // exact for duplicate-free and duplicate cases alike
-      atomicAdd(&dx[t * 2 * feat2 + 2 * f], gr.x);
-      atomicAdd(&dx[t * 2 * feat2 + 2 * f + 1], gr.y);
-    }
-    if (e <= s) continue;
-    const float inv = 1.0f / (float)(e - s);
-    for (int64_t f = lane; f < feat2; f += kWave) {
-      const float2 g = __bfloat1622float2(dy[t * ostride + f]);
-      const float gx = g.x * inv, gy = g.y * inv;
-      for (int64_t k = s; k < e; ++k) {
-        const int64_t c = col[k];
-        atomicAdd(&dx[c * 2 * feat2 + 2 * f], gx);
-        atomicAdd(&dx[c * 2 * feat2 + 2 * f + 1], gy);
-      }
-    }
-  }
-}
-
 int wave_grid(int64_t rows) {
   const int64_t waves = std::min<int64_t>(rows, (int64_t)kMaxBlocks * 4);
   return (int)std::min<int64_t>((waves * kWave + kBlock - 1) / kBlock,
@@ -345,17 +287,6 @@ torch::Tensor hip_segment_mean_cat_bwd(const torch::Tensor& dy,
   auto dx = torch::zeros({n_src, feat}, dy.options().dtype(torch::kFloat32));
   if (n_tgt > 0) {
     auto dyc = dy.contiguous();
-    if (dy.scalar_type() == torch::kBFloat16 && feat % 2 == 0) {
-      hipLaunchKernelGGL(seg_mean_cat_bwd_bf162_kernel,
-                         dim3(wave_grid(n_tgt)), dim3(kBlock), 0,
-                         current_stream(),
-                         reinterpret_cast<const __hip_bfloat162*>(
-                             dyc.data_ptr()),
-                         col.data_ptr<int64_t>(),
-                         offsets.data_ptr<int64_t>(), n_tgt, feat / 2,
-                         dx.data_ptr<float>());
-      return dx;
-    }
     GLT_DISPATCH_SEG(dy.scalar_type(), "segment_mean_cat_bwd", [&] {
       hipLaunchKernelGGL(seg_mean_cat_bwd_kernel<scalar_t>,
                          dim3(wave_grid(n_tgt)), dim3(kBlock), 0,
@@ -377,17 +308,6 @@ torch::Tensor hip_segment_mean_bwd(const torch::Tensor& dy,
   auto dx = torch::zeros({n_src, feat}, dy.options().dtype(torch::kFloat32));
   if (n_tgt > 0) {
     auto dyc = dy.contiguous();
-    if (dy.scalar_type() == torch::kBFloat16 && feat % 2 == 0) {
-      hipLaunchKernelGGL(seg_mean_bwd_bf162_kernel,
-                         dim3(wave_grid(n_tgt)), dim3(kBlock), 0,
-                         current_stream(),
-                         reinterpret_cast<const __hip_bfloat162*>(
-                             dyc.data_ptr()),
-                         col.data_ptr<int64_t>(),
-                         offsets.data_ptr<int64_t>(), n_tgt, feat / 2,
-                         dx.data_ptr<float>());
-      return dx;
-    }
     GLT_DISPATCH_SEG(dy.scalar_type(), "segment_mean_bwd", [&] {
       hipLaunchKernelGGL(seg_mean_bwd_kernel<scalar_t>,
                          dim3(wave_grid(n_tgt)), dim3(kBlock), 0,
