@@ -67,16 +67,22 @@ class DistNeighborSampler:
                  device: Optional[torch.device] = None,
                  concurrency: int = 4,
                  channel: Optional[ChannelBase] = None,
-                 use_all2all: bool = False):
+                 use_all2all: bool = False,
+                 all2all_group=None):
         """use_all2all: collect remote feature rows with
         torch.distributed.all_to_all_single (RCCL over xGMI intra-node)
         instead of per-partition RPC pulls.  Requires one rank per
-        partition stepping in LOCKSTEP (same batch count, shuffle
-        deterministic) and forces concurrency=1 so the collectives stay
+        partition (of `all2all_group`, default group) stepping in
+        LOCKSTEP and forces concurrency=1 so the collectives stay
         ordered (parity: the reference's optional gloo use_all2all path,
-        reference dist_sampling_producer.py:73-80)."""
+        reference dist_sampling_producer.py:73-80).  The mp producers
+        enforce lockstep with a per-epoch batch-count handshake:
+        `set_all2all_budget(n)` caps the number of collective batches;
+        the ragged tail falls back to the RPC pull path."""
         self.data = data
         self.use_all2all = use_all2all
+        self.all2all_group = all2all_group
+        self._a2a_budget: Optional[int] = None  # None = unlimited
         if use_all2all:
             concurrency = 1
         self.num_neighbors = num_neighbors
@@ -483,18 +489,37 @@ class DistNeighborSampler:
         return encode_sampler_output(out, x=x or None, y=y or None,
                                      edge_attr=ea or None)
 
+    def set_all2all_budget(self, n: Optional[int]):
+        """Number of upcoming batches allowed to use the collective path
+        (must be set identically on every rank of the group); None =
+        unlimited.  Consumed once per homo node-feature collection."""
+        self._a2a_budget = n
+
+    def _a2a_ready(self) -> bool:
+        import torch.distributed as dist
+
+        if not dist.is_initialized():
+            return False
+        if dist.get_world_size(self.all2all_group) != \
+                self.data.num_partitions:
+            return False
+        if self._a2a_budget is None:
+            return True
+        if self._a2a_budget <= 0:
+            return False
+        self._a2a_budget -= 1
+        return True
+
     async def _collect_one(self, loop, kind, ids, type_key):
         # all2all only for the homo path: hetero collection gates each
         # (kind, type) on per-rank conditions (nodes.numel(), has()), so
         # ranks could issue different numbers of all_to_all_single calls
         # and deadlock the RCCL group — force the RPC pull path there.
-        if self.use_all2all and type_key is None and kind == "node":
-            import torch.distributed as dist
-
-            if dist.is_initialized() and \
-                    dist.get_world_size() == self.data.num_partitions:
-                return self.dist_feature.all2all_get(
-                    kind, ids.cpu(), type_key).to(self.sampler.device)
+        if self.use_all2all and type_key is None and kind == "node" \
+                and self._a2a_ready():
+            return self.dist_feature.all2all_get(
+                kind, ids.cpu(), type_key,
+                group=self.all2all_group).to(self.sampler.device)
         futures, positions, local_vals, local_pos = \
             self.dist_feature.async_get(kind, ids, type_key)
         if not futures:

@@ -57,13 +57,24 @@ class MpDistSamplingWorkerOptions(_BasicDistSamplingWorkerOptions):
                  worker_concurrency: int = 4, master_addr=None,
                  master_port=None, num_rpc_threads: int = 16,
                  rpc_timeout: float = 180.0, channel_size="256MB",
-                 channel_capacity: int = 128, pin_memory: bool = False):
+                 channel_capacity: int = 128, pin_memory: bool = False,
+                 use_all2all: bool = True,
+                 a2a_port: Optional[int] = None):
+        """use_all2all (default on): sampling workers exchange feature
+        rows with torch.distributed all_to_all collectives instead of
+        per-partition RPC pulls when the topology allows (one rank per
+        partition, homo node sampling with feature collection).  Workers
+        rendezvous their own process group on `a2a_port` (default
+        master_port + 1) and enforce lockstep with a per-epoch
+        batch-count handshake; the ragged tail and every unsupported
+        shape fall back to the RPC path automatically."""
         super().__init__(num_workers, worker_devices, worker_concurrency,
                          master_addr, master_port, num_rpc_threads,
-                         rpc_timeout)
+                         rpc_timeout, use_all2all)
         self.channel_size = channel_size
         self.channel_capacity = channel_capacity
         self.pin_memory = pin_memory
+        self.a2a_port = a2a_port
 
 
 class RemoteDistSamplingWorkerOptions(_BasicDistSamplingWorkerOptions):

@@ -50,13 +50,49 @@ def _sampling_worker_loop(worker_idx: int, dataset: DistDataset,
                  worker_options.master_port,
                  worker_options.num_rpc_threads,
                  worker_options.rpc_timeout)
+        # optional all2all transport: sampling workers form their own
+        # torch.distributed world (gloo — feature rows are served from
+        # host memory); worker i of every trainer makes a subgroup, so
+        # collectives pair partition-to-partition at equal worker index.
+        a2a_group = None
+        use_a2a = (getattr(worker_options, "use_all2all", False)
+                   and world_size > 1
+                   and worker_options.master_port is not None
+                   and sampling_config.sampling_type == SamplingType.NODE
+                   and sampling_config.collect_features
+                   and not isinstance(dataset.graph, dict)
+                   and dataset.num_partitions == world_size)
+        if use_a2a:
+            import datetime
+
+            import torch.distributed as dist
+
+            a2a_port = getattr(worker_options, "a2a_port", None) or \
+                int(worker_options.master_port) + 1
+            try:
+                if not dist.is_initialized():
+                    dist.init_process_group(
+                        "gloo",
+                        init_method="tcp://%s:%d" % (
+                            worker_options.master_addr, a2a_port),
+                        world_size=world_size * num_workers,
+                        rank=rank * num_workers + worker_idx,
+                        timeout=datetime.timedelta(seconds=120))
+                groups = [dist.new_group([p * num_workers + i
+                                          for p in range(world_size)])
+                          for i in range(num_workers)]
+                a2a_group = groups[worker_idx]
+            except Exception:  # rendezvous failed: pure-RPC fallback
+                a2a_group = None
+                use_a2a = False
         sampler = DistNeighborSampler(
             dataset, sampling_config.num_neighbors,
             with_edge=sampling_config.with_edge,
             with_weight=sampling_config.with_weight,
             collect_features=sampling_config.collect_features,
             edge_dir=sampling_config.edge_dir, device=device,
-            concurrency=worker_options.worker_concurrency, channel=channel)
+            concurrency=worker_options.worker_concurrency, channel=channel,
+            use_all2all=use_a2a, all2all_group=a2a_group)
         while True:
             cmd = task_queue.get()
             if cmd == _Cmd.STOP:
@@ -64,6 +100,18 @@ def _sampling_worker_loop(worker_idx: int, dataset: DistDataset,
             # sample the whole assigned shard for this epoch
             bs = sampling_config.batch_size
             n = len(seeds_input)
+            if a2a_group is not None:
+                # lockstep handshake: collective batches = the minimum
+                # epoch batch count across the subgroup; the tail RPCs
+                import torch.distributed as dist
+
+                nb = (n // bs) if sampling_config.drop_last else \
+                    (n + bs - 1) // bs
+                cnt = torch.tensor([nb])
+                outs = [torch.zeros_like(cnt) for _ in range(world_size)]
+                dist.all_gather(outs, cnt, group=a2a_group)
+                sampler.set_all2all_budget(
+                    min(int(t.item()) for t in outs))
             order = torch.randperm(n) if sampling_config.shuffle else \
                 torch.arange(n)
             futures = []
@@ -99,6 +147,13 @@ def _sampling_worker_loop(worker_idx: int, dataset: DistDataset,
                       f"{len(errors)} batch(es) failed:\n{errors[0]}",
                       file=sys.stderr, flush=True)
         sampler.shutdown()
+        if a2a_group is not None:
+            import torch.distributed as dist
+
+            try:
+                dist.destroy_process_group()
+            except Exception:
+                pass
         shutdown_rpc()
     except KeyboardInterrupt:
         pass

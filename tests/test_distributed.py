@@ -800,3 +800,63 @@ def test_dist_hetero_link_loader_remote_fanout():
             p.terminate()
     for rank, err in results:
         assert err is None, f"rank {rank}:\n{err}"
+
+
+def _worker_mp_a2a_ragged(rank, world, port, fail_q):
+    try:
+        import glt_amd
+        from glt_amd.distributed import (MpDistSamplingWorkerOptions,
+                                         DistNeighborLoader,
+                                         init_worker_group)
+
+        glt_amd.seed_everything(31 + rank)
+        init_worker_group(world, rank)
+        ds = _build_partition(rank)
+        opts = MpDistSamplingWorkerOptions(
+            num_workers=1, master_addr="127.0.0.1", master_port=port,
+            channel_size="16MB", channel_capacity=16,
+            use_all2all=True)
+        # ragged: rank 0 seeds 4 batches, rank 1 only 3 — the lockstep
+        # handshake must cap the collective path at 3 and serve rank 0's
+        # tail batch over RPC
+        n_seed = 20 if rank == 0 else 15
+        seeds = torch.arange(rank, VNUM, 2)[:n_seed // 1][: (20 if rank == 0
+                                                             else 15)]
+        loader = DistNeighborLoader(ds, [2, 2], input_nodes=seeds,
+                                    batch_size=5, shuffle=False,
+                                    worker_options=opts)
+        for epoch in range(2):
+            n = 0
+            for data in loader:
+                _check_batch(data)
+                n += 1
+            assert n == (4 if rank == 0 else 3), n
+        loader.shutdown()
+        fail_q.put((rank, None))
+    except Exception:  # noqa: BLE001
+        import traceback
+
+        fail_q.put((rank, traceback.format_exc()))
+
+
+@pytest.mark.timeout(300)
+def test_dist_mp_all2all_ragged_tail():
+    """all2all default mp producers with UNEQUAL per-rank batch counts:
+    the per-epoch batch-count handshake must keep the collective calls
+    matched and RPC-serve the ragged tail (docs/round2_designs.md §4)."""
+    from glt_amd.utils import get_free_port
+
+    ctx = mp.get_context("spawn")
+    port = get_free_port()
+    fail_q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_mp_a2a_ragged,
+                         args=(r, 2, port, fail_q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = [fail_q.get(timeout=280) for _ in range(2)]
+    for p in procs:
+        p.join(timeout=30)
+        if p.is_alive():
+            p.terminate()
+    for rank, err in results:
+        assert err is None, f"rank {rank}:\n{err}"
