@@ -95,8 +95,14 @@ class Feature:
         groups = self.device_group_list
         if hot > 0:
             if groups:
-                # Shard hot rows across the group's devices (xGMI peers).
-                devs = groups[0].device_list
+                # Shard hot rows across the devices of the group that
+                # contains `dev` (hot-tier REPLICATED per DeviceGroup,
+                # sharded within it — parity: reference
+                # python/data/feature.py:89-141; round-1 used only
+                # group[0] regardless of the gathering device)
+                group = next((g for g in groups
+                              if dev in g.device_list), groups[0])
+                devs = group.device_list
                 per = (hot + len(devs) - 1) // len(devs)
                 start = 0
                 for d in devs:

@@ -409,7 +409,7 @@ def _ipc_graph_child(g, q):
         assert g._lazy_done and g.indptr.is_cuda  # no re-materialization
         seeds = torch.arange(10, device="cuda")
         nbrs, num, _ = _C.sample_neighbors(g.indptr, g.indices, seeds, 2)
-        d = (nbrs.cpu() - seeds.repeat_interleave(num.cpu()).cpu()) % 1000
+        d = (nbrs.cpu() - seeds.repeat_interleave(num).cpu()) % 1000
         assert ((d == 1) | (d == 2)).all()
         q.put((None, None))
     except Exception:  # noqa: BLE001
