@@ -15,12 +15,16 @@ def init_client(num_servers: int, num_clients: int, client_rank: int,
                 master_addr: str, master_port: int,
                 num_rpc_threads: int = 4, rpc_timeout: float = 240.0,
                 client_group_name: str = "distributed_client",
-                is_dynamic: bool = False):
+                is_dynamic: bool = False,
+                server_group_name: str = "distributed_server"):
+    """is_dynamic worlds resolve peer names through the registry living
+    on `{server_group_name}_0` (the rendezvous anchor)."""
     global _assigned_servers
     _set_client_context(num_servers, num_clients, client_rank,
                         client_group_name)
     init_rpc(master_addr, master_port, num_rpc_threads, rpc_timeout,
-             is_dynamic=is_dynamic)
+             is_dynamic=is_dynamic,
+             anchor_name=f"{server_group_name}_0" if is_dynamic else None)
     _assigned_servers = assign_server_by_order(client_rank, num_servers,
                                                num_clients)
 
@@ -30,7 +34,10 @@ def get_assigned_servers() -> List[int]:
 
 
 def _server_name(server_rank: int) -> str:
-    return f"distributed_server_{server_rank}"
+    from .dist_context import DistRole
+    from .rpc import group_worker_name
+
+    return group_worker_name(server_rank, DistRole.SERVER)
 
 
 def request_server(server_rank: int, func_name: str, *args, **kwargs):
