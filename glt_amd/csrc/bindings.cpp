@@ -22,12 +22,13 @@ using OptTensor = c10::optional<torch::Tensor>;
 std::tuple<torch::Tensor, torch::Tensor, OptTensor> sample_neighbors(
     const torch::Tensor& indptr, const torch::Tensor& indices,
     const OptTensor& edge_ids, const OptTensor& edge_weights,
-    const torch::Tensor& seeds, int64_t k, bool with_edge, bool weighted) {
+    const torch::Tensor& seeds, int64_t k, bool with_edge, bool weighted,
+    bool replace) {
   if (seeds.is_cuda())
     return hip_sample_neighbors(indptr, indices, edge_ids, edge_weights,
-                                seeds, k, with_edge, weighted);
+                                seeds, k, with_edge, weighted, replace);
   return cpu_sample_neighbors(indptr, indices, edge_ids, edge_weights, seeds,
-                              k, with_edge, weighted);
+                              k, with_edge, weighted, replace);
 }
 
 torch::Tensor lookup_degree(const torch::Tensor& indptr,
@@ -149,13 +150,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       "sample_neighbors",
       [](const torch::Tensor& indptr, const torch::Tensor& indices,
          const torch::Tensor& seeds, int64_t k, const OptTensor& edge_ids,
-         const OptTensor& edge_weights, bool with_edge, bool weighted) {
+         const OptTensor& edge_weights, bool with_edge, bool weighted,
+         bool replace) {
         return sample_neighbors(indptr, indices, edge_ids, edge_weights,
-                                seeds, k, with_edge, weighted);
+                                seeds, k, with_edge, weighted, replace);
       },
       py::arg("indptr"), py::arg("indices"), py::arg("seeds"), py::arg("k"),
       py::arg("edge_ids") = py::none(), py::arg("edge_weights") = py::none(),
-      py::arg("with_edge") = false, py::arg("weighted") = false);
+      py::arg("with_edge") = false, py::arg("weighted") = false,
+      py::arg("replace") = true);
   m.def("lookup_degree", &lookup_degree);
   m.def("sample_negative", &sample_negative, py::arg("indptr"),
         py::arg("indices"), py::arg("num_cols"), py::arg("req_num"),

@@ -70,7 +70,7 @@ cpu_sample_neighbors(const torch::Tensor& indptr, const torch::Tensor& indices,
                      const c10::optional<torch::Tensor>& edge_ids,
                      const c10::optional<torch::Tensor>& edge_weights,
                      const torch::Tensor& seeds, int64_t k, bool with_edge,
-                     bool weighted) {
+                     bool weighted, bool replace) {
   check_int64_1d(indptr, "indptr");
   check_int64_1d(indices, "indices");
   check_int64_1d(seeds, "seeds");
@@ -125,7 +125,7 @@ cpu_sample_neighbors(const torch::Tensor& indptr, const torch::Tensor& indices,
           out[j] = view.indices[base + pick[j]];
           if (oe) oe[j] = view.eids[base + pick[j]];
         }
-      } else {
+      } else if (replace) {
         // CDF + binary search, with replacement.
         pick.resize(deg);
         double acc = 0.0;
@@ -139,6 +139,32 @@ cpu_sample_neighbors(const torch::Tensor& indptr, const torch::Tensor& indices,
           int64_t idx = std::lower_bound(cdf.begin(), cdf.end(), r) -
                         cdf.begin();
           if (idx >= deg) idx = deg - 1;
+          out[j] = view.indices[base + idx];
+          if (oe) oe[j] = view.eids[base + idx];
+        }
+      } else {
+        // Weighted WITHOUT replacement: Efraimidis-Spirakis exponential
+        // race — the kk smallest keys -log(u_i)/w_i (matches the GPU
+        // kernel; zero/negative weights never selected, uniform pad if
+        // fewer than kk positive weights).
+        std::vector<std::pair<float, int64_t>> keys;
+        keys.reserve(deg);
+        for (int64_t j = 0; j < deg; ++j) {
+          const float w = view.weights[base + j];
+          if (!(w > 0.f)) continue;
+          float u = rng.uniform_float();
+          u = u < 1e-12f ? 1e-12f : u;
+          keys.emplace_back(-std::log(u) / w, j);
+        }
+        const int64_t npos = (int64_t)keys.size();
+        const int64_t take = std::min(npos, kk);
+        std::partial_sort(keys.begin(), keys.begin() + take, keys.end());
+        for (int64_t j = 0; j < take; ++j) {
+          out[j] = view.indices[base + keys[j].second];
+          if (oe) oe[j] = view.eids[base + keys[j].second];
+        }
+        for (int64_t j = take; j < kk; ++j) {
+          const int64_t idx = (int64_t)rng.uniform((uint64_t)deg);
           out[j] = view.indices[base + idx];
           if (oe) oe[j] = view.eids[base + idx];
         }

@@ -13,7 +13,7 @@ cpu_sample_neighbors(const torch::Tensor& indptr, const torch::Tensor& indices,
                      const c10::optional<torch::Tensor>& edge_ids,
                      const c10::optional<torch::Tensor>& edge_weights,
                      const torch::Tensor& seeds, int64_t k, bool with_edge,
-                     bool weighted);
+                     bool weighted, bool replace = true);
 torch::Tensor cpu_lookup_degree(const torch::Tensor& indptr,
                                 const torch::Tensor& nodes);
 torch::Tensor cpu_cal_nbr_prob(const torch::Tensor& indptr,

@@ -85,7 +85,8 @@ class NeighborSampler(BaseSampler):
                  device: Optional[torch.device] = None,
                  with_edge: bool = False, with_neg: bool = False,
                  with_weight: bool = False, edge_dir: str = "out",
-                 seed: Optional[int] = None):
+                 seed: Optional[int] = None,
+                 weight_replace: bool = True):
         from .. import _C
 
         self._C = _C
@@ -94,6 +95,9 @@ class NeighborSampler(BaseSampler):
         self.with_edge = with_edge
         self.with_neg = with_neg
         self.with_weight = with_weight
+        # weighted draws: with replacement (reference semantics) or
+        # Efraimidis-Spirakis without replacement (GPU+CPU kernels)
+        self.weight_replace = weight_replace
         self.edge_dir = edge_dir
         self.is_hetero = isinstance(graph, dict)
         self._lock = threading.Lock()
@@ -170,7 +174,8 @@ class NeighborSampler(BaseSampler):
             g.indptr, g.indices, srcs, k,
             edge_ids=g.edge_ids if self.with_edge else None,
             edge_weights=g.edge_weights if weighted else None,
-            with_edge=self.with_edge, weighted=weighted)
+            with_edge=self.with_edge, weighted=weighted,
+            replace=getattr(self, "weight_replace", True))
         return NeighborOutput(nbrs, num, eids)
 
     # -- homo ----------------------------------------------------------

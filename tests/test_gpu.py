@@ -871,3 +871,54 @@ def test_feature_multi_device_group_selection():
     assert len(f._keepalive) == 1  # one shard: group of device 0 only
     ids = torch.randint(0, 1000, (256,))
     assert torch.equal(f[ids].cpu(), feats[ids])
+
+
+def test_weighted_sampler_hub_row_gpu():
+    """10k-degree hub row: CDF build + binary-search draws must stay
+    correct and proportionally biased (round-1 scanned the CDF linearly
+    per draw — O(k*deg))."""
+    glt_amd.seed_everything(11)
+    deg = 10_000
+    rows = torch.zeros(deg, dtype=torch.long)
+    cols = torch.arange(1, deg + 1)
+    # one heavy neighbor (id 1) with half the total mass
+    w = torch.ones(deg)
+    w[0] = float(deg - 1)
+    topo = Topology(torch.stack([rows, cols]), edge_weights=w,
+                    num_nodes=deg + 1)
+    seeds = torch.zeros(2000, dtype=torch.long, device="cuda")
+    nbrs, num, _ = _C.sample_neighbors(
+        topo.indptr.cuda(), topo.indices.cuda(), seeds, 4,
+        edge_weights=topo.edge_weights.cuda(), weighted=True)
+    assert (num == 4).all()
+    vals = nbrs.cpu()
+    assert ((vals >= 1) & (vals <= deg)).all()
+    frac_heavy = (vals == 1).float().mean().item()
+    assert 0.42 < frac_heavy < 0.58, frac_heavy  # ~1/2 the mass
+
+
+def test_weighted_sampler_no_replace_gpu():
+    """Without-replacement weighted draws: distinct per row, biased, and
+    CPU/GPU agreement on the heavy-element inclusion rate."""
+    glt_amd.seed_everything(12)
+    deg, k = 500, 8
+    rows = torch.zeros(deg, dtype=torch.long)
+    cols = torch.arange(1, deg + 1)
+    w = torch.ones(deg)
+    w[:4] = 200.0  # 4 heavy neighbors: ids 1..4 (after row sort)
+    topo = Topology(torch.stack([rows, cols]), edge_weights=w,
+                    num_nodes=deg + 1)
+    seeds = torch.zeros(1000, dtype=torch.long, device="cuda")
+    nbrs, num, _ = _C.sample_neighbors(
+        topo.indptr.cuda(), topo.indices.cuda(), seeds, k,
+        edge_weights=topo.edge_weights.cuda(), weighted=True,
+        replace=False)
+    assert (num == k).all()
+    vals = nbrs.cpu().view(-1, k)
+    for r in range(0, 1000, 97):
+        row = vals[r].tolist()
+        assert len(set(row)) == k, "duplicates in no-replace draw"
+    # each heavy id should be present in nearly every draw
+    heavy_rate = sum((vals == i).any(1).float().mean().item()
+                     for i in (1, 2, 3, 4)) / 4
+    assert heavy_rate > 0.9, heavy_rate

@@ -222,3 +222,33 @@ def test_sampling_reproducible_under_seed(ring_graph):
         assert torch.equal(x.node, y.node)
         assert torch.equal(x.row, y.row)
         assert torch.equal(x.col, y.col)
+
+
+def test_weighted_no_replace_cpu():
+    """CPU Efraimidis-Spirakis without-replacement: distinct, biased,
+    zero-weight exclusion, short-row pad."""
+    import glt_amd
+    from glt_amd import _C
+    from glt_amd.data import Topology
+
+    glt_amd.seed_everything(4)
+    deg, k = 100, 10
+    rows = torch.zeros(deg, dtype=torch.long)
+    cols = torch.arange(1, deg + 1)
+    w = torch.ones(deg)
+    w[:3] = 500.0
+    w[50:] = 0.0  # zero weights never selected (enough positives remain)
+    topo = Topology(torch.stack([rows, cols]), edge_weights=w,
+                    num_nodes=deg + 1)
+    seeds = torch.zeros(300, dtype=torch.long)
+    nbrs, num, _ = _C.sample_neighbors(
+        topo.indptr, topo.indices, seeds, k,
+        edge_weights=topo.edge_weights, weighted=True, replace=False)
+    assert (num == k).all()
+    vals = nbrs.view(-1, k)
+    assert (vals <= 50).all()  # zero-weight tail excluded
+    for r in range(0, 300, 29):
+        assert len(set(vals[r].tolist())) == k
+    heavy_rate = sum((vals == i).any(1).float().mean().item()
+                     for i in (1, 2, 3)) / 3
+    assert heavy_rate > 0.95, heavy_rate

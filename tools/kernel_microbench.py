@@ -51,6 +51,22 @@ def build_flagship_shapes(op, device):
         indices = torch.randint(0, n, (n * deg,), device=device)
         seeds = torch.randint(0, n, (131_072,), device=device)
         return (indptr, indices, seeds)
+    if op in ("sample_weighted", "sample_weighted_nr"):
+        n, deg = 500_000, 50
+        indptr = torch.arange(0, (n + 1) * deg, deg, device=device)
+        indices = torch.randint(0, n, (n * deg,), device=device)
+        w = torch.rand(n * deg, device=device)
+        seeds = torch.randint(0, n, (131_072,), device=device)
+        return (indptr, indices, w, seeds)
+    if op == "sample_weighted_hub":
+        # power-law-ish: 1000 hub rows of degree 10k + filler rows
+        n_hub, hub_deg = 1000, 10_000
+        n = n_hub
+        indptr = torch.arange(0, (n + 1) * hub_deg, hub_deg, device=device)
+        indices = torch.randint(0, 1 << 20, (n * hub_deg,), device=device)
+        w = torch.rand(n * hub_deg, device=device)
+        seeds = torch.randint(0, n, (8192,), device=device)
+        return (indptr, indices, w, seeds)
     raise SystemExit(f"unknown op {op}")
 
 
@@ -58,7 +74,9 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--op", required=True,
                     choices=["seg_mean", "seg_mean_cat", "gat_fused",
-                             "mfma_gemm", "gather", "sample"])
+                             "mfma_gemm", "gather", "sample",
+                             "sample_weighted", "sample_weighted_nr",
+                             "sample_weighted_hub"])
     ap.add_argument("--iters", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=3)
     args = ap.parse_args()
@@ -69,6 +87,7 @@ def main():
 
     inp = build_flagship_shapes(args.op, device)
 
+    op = args.op
     if args.op == "seg_mean":
         from glt_amd.ops.segment import _boundaries
         x, tgt, src, n_tgt = inp
@@ -92,9 +111,15 @@ def main():
         store = _C.UnifiedFeatureStore(0)
         store.append(feats)
         fn = lambda: store.gather(rows)
-    else:  # sample
+    elif op == "sample":
         indptr, indices, seeds = inp
         fn = lambda: _C.sample_neighbors(indptr, indices, seeds, 15)
+    else:  # weighted variants
+        indptr, indices, w, seeds = inp
+        rep = op != "sample_weighted_nr"
+        fn = lambda: _C.sample_neighbors(indptr, indices, seeds, 15,
+                                         edge_weights=w, weighted=True,
+                                         replace=rep)
 
     for _ in range(args.warmup):
         fn()
