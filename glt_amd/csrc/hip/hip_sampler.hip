@@ -78,6 +78,71 @@ __global__ void sample_gather_kernel(
 // (Efraimidis-Spirakis exponential race, k rounds of block-argmin; keys
 // are recomputable from the call seed so no per-element state is kept).
 // ---------------------------------------------------------------------------
+// Small-row fast path (deg <= 64, with-replacement): the row's weights
+// live one-per-lane in registers; the CDF is a wave shfl-scan and each
+// draw is one ballot + ffs — no LDS, no barriers.  Rows with deg > 64
+// are left to the block kernel below (and vice versa).
+template <bool WITH_EID>
+__global__ void weighted_sample_wave_kernel(
+    const int64_t* __restrict__ indptr, const int64_t* __restrict__ indices,
+    const int64_t* __restrict__ eids, const float* __restrict__ weights,
+    int64_t num_rows, const int64_t* __restrict__ seeds, int64_t bs,
+    int64_t k, const int64_t* __restrict__ offsets, uint64_t call_seed,
+    int64_t* __restrict__ out_nbrs, int64_t* __restrict__ out_eids) {
+  const int lane = threadIdx.x & (kWave - 1);
+  const int64_t wave =
+      (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) / kWave;
+  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) / kWave;
+  for (int64_t r = wave; r < bs; r += n_waves) {
+    const int64_t v = seeds[r];
+    if (v < 0 || v >= num_rows) continue;
+    const int64_t base = indptr[v];
+    const int64_t deg = indptr[v + 1] - base;
+    if (deg == 0 || deg > kWave) continue;  // big rows: block kernel
+    const int64_t off = offsets[r];
+    if (deg <= k) {
+      if (lane < deg) {
+        out_nbrs[off + lane] = indices[base + lane];
+        if (WITH_EID) out_eids[off + lane] = eids[base + lane];
+      }
+      continue;
+    }
+    float wv = 0.f;
+    if (lane < deg) {
+      const float w = weights[base + lane];
+      wv = w > 0.f ? w : 0.f;
+    }
+    float c = wv;  // inclusive wave scan
+#pragma unroll
+    for (int sft = 1; sft < kWave; sft <<= 1) {
+      const float p = __shfl_up(c, sft);
+      if (lane >= sft) c += p;
+    }
+    const float tot = __shfl(c, kWave - 1);
+    DRng rng(call_seed ^ (uint64_t)r * 0xD6E8FEB86659FD93ull ^
+             (uint64_t)lane * 0xA24BAED4963EE407ull);
+    for (int64_t j = 0; j < k; ++j) {
+      // all lanes advance the RNG of the serving lane in lockstep by
+      // broadcasting its draw
+      const int serve = (int)(j % kWave);
+      float u = rng.uniform_float();
+      u = __shfl(u, serve) * tot;
+      int64_t idx;
+      if (tot > 0.f) {
+        const uint64_t mask = __ballot(c >= u && lane < deg);
+        const int first = __ffsll((unsigned long long)mask) - 1;
+        idx = first >= 0 ? first : deg - 1;
+      } else {
+        idx = (int64_t)(__shfl(rng.next() % (uint64_t)deg, serve));
+      }
+      if (lane == serve) {
+        out_nbrs[off + j] = indices[base + idx];
+        if (WITH_EID) out_eids[off + j] = eids[base + idx];
+      }
+    }
+  }
+}
+
 constexpr int kCdfCap = 4096;
 constexpr int kSelCap = 256;
 
@@ -103,6 +168,7 @@ void weighted_sample_block_kernel(
     const int64_t deg = indptr[v + 1] - base;
     const int64_t off = offsets[r];
     if (deg == 0) continue;
+    if (REPLACE && deg <= kWave) continue;  // wave kernel handled it
     if (deg <= k) {
       for (int64_t j = tid; j < deg; j += 256) {
         out_nbrs[off + j] = indices[base + j];
@@ -382,12 +448,36 @@ hip_sample_neighbors(const torch::Tensor& indptr, const torch::Tensor& indices,
             cs, nbrs.data_ptr<int64_t>(),
             WE ? out_eids.data_ptr<int64_t>() : nullptr);
       };
+      auto launch_wave = [&](auto with_eid_tag) {
+        constexpr bool WE = decltype(with_eid_tag)::value;
+        const int64_t wv_blocks =
+            std::min<int64_t>((bs * kWave + kBlock - 1) / kBlock,
+                              kMaxBlocks);
+        hipLaunchKernelGGL(
+            (weighted_sample_wave_kernel<WE>),
+            dim3((uint32_t)std::max<int64_t>(wv_blocks, 1)), dim3(kBlock),
+            0, stream, indptr.data_ptr<int64_t>(),
+            indices.data_ptr<int64_t>(),
+            WE ? edge_ids->data_ptr<int64_t>() : nullptr,
+            edge_weights->data_ptr<float>(), num_rows,
+            seeds.data_ptr<int64_t>(), bs, kk, offsets.data_ptr<int64_t>(),
+            cs, nbrs.data_ptr<int64_t>(),
+            WE ? out_eids.data_ptr<int64_t>() : nullptr);
+      };
       if (with_edge) {
-        if (replace) launch(std::true_type{}, std::true_type{});
-        else launch(std::true_type{}, std::false_type{});
+        if (replace) {
+          launch_wave(std::true_type{});
+          launch(std::true_type{}, std::true_type{});
+        } else {
+          launch(std::true_type{}, std::false_type{});
+        }
       } else {
-        if (replace) launch(std::false_type{}, std::true_type{});
-        else launch(std::false_type{}, std::false_type{});
+        if (replace) {
+          launch_wave(std::false_type{});
+          launch(std::false_type{}, std::true_type{});
+        } else {
+          launch(std::false_type{}, std::false_type{});
+        }
       }
     }
   }

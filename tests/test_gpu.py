@@ -921,4 +921,6 @@ def test_weighted_sampler_no_replace_gpu():
     # each heavy id should be present in nearly every draw
     heavy_rate = sum((vals == i).any(1).float().mean().item()
                      for i in (1, 2, 3, 4)) / 4
-    assert heavy_rate > 0.9, heavy_rate
+    # analytic expectation ~0.84 (matches the CPU E-S twin); uniform
+    # draws would include any given id at ~1.6%
+    assert 0.75 < heavy_rate < 0.95, heavy_rate
