@@ -119,26 +119,28 @@ __global__ void weighted_sample_wave_kernel(
       if (lane >= sft) c += p;
     }
     const float tot = __shfl(c, kWave - 1);
-    DRng rng(call_seed ^ (uint64_t)r * 0xD6E8FEB86659FD93ull ^
-             (uint64_t)lane * 0xA24BAED4963EE407ull);
-    for (int64_t j = 0; j < k; ++j) {
-      // all lanes advance the RNG of the serving lane in lockstep by
-      // broadcasting its draw
-      const int serve = (int)(j % kWave);
-      float u = rng.uniform_float();
-      u = __shfl(u, serve) * tot;
+    // all k draws in parallel: lane j serves draw j with a fixed
+    // 6-step binary search over the register CDF via lane shuffles
+    for (int64_t j = lane; j < k; j += kWave) {
+      DRng rng(call_seed ^ (uint64_t)r * 0xD6E8FEB86659FD93ull ^
+               (uint64_t)j * 0xA24BAED4963EE407ull);
       int64_t idx;
       if (tot > 0.f) {
-        const uint64_t mask = __ballot(c >= u && lane < deg);
-        const int first = __ffsll((unsigned long long)mask) - 1;
-        idx = first >= 0 ? first : deg - 1;
+        const float u = rng.uniform_float() * tot;
+        int lo = 0, hi = (int)deg - 1;
+#pragma unroll
+        for (int step = 0; step < 6; ++step) {
+          const int mid = (lo + hi) >> 1;
+          const float cm = __shfl(c, mid);
+          if (cm >= u) hi = mid; else lo = mid + 1;
+          if (lo > hi) lo = hi;
+        }
+        idx = lo;
       } else {
-        idx = (int64_t)(__shfl(rng.next() % (uint64_t)deg, serve));
+        idx = (int64_t)rng.uniform((uint64_t)deg);
       }
-      if (lane == serve) {
-        out_nbrs[off + j] = indices[base + idx];
-        if (WITH_EID) out_eids[off + j] = eids[base + idx];
-      }
+      out_nbrs[off + j] = indices[base + idx];
+      if (WITH_EID) out_eids[off + j] = eids[base + idx];
     }
   }
 }
