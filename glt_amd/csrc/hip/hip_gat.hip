@@ -68,26 +68,48 @@ __global__ void gat_fused_fwd_kernel(
         (kWave + lane < C ? tv[kWave + lane] * atd[kWave + lane] : 0.f));
     float m = -1e30f, Z = 0.f;
     float acc0 = 0.f, acc1 = 0.f;  // lanes cover C (up to 2 passes)
-    for (int64_t e = s0; e < s1; ++e) {
-      const int64_t sn = src[e];
-      const float* hv = h_src + (sn * H + h) * C;
-      const float h0 = lane < C ? hv[lane] : 0.f;
-      const float h1 = kWave + lane < C ? hv[kWave + lane] : 0.f;
-      float s = ad + wave_sum(h0 * as0 + h1 * as1);
-      if (lane == 0) spre_out[e * H + h] = s;  // backward reuses this
-      s = s > 0.f ? s : s * slope;
-      float scale = 1.f;
-      float p;
-      if (s > m) {
-        scale = __expf(m - s);
-        p = 1.f;
-        m = s;
-      } else {
-        p = __expf(s - m);
+    // 4-edge load batching (see backward): the online-softmax chain
+    // stays serial but its global loads overlap.
+    for (int64_t e = s0; e < s1; e += 4) {
+      const int nb = (int)((s1 - e) < 4 ? (s1 - e) : 4);
+      int64_t sn[4];
+      float hh0[4], hh1[4];
+#pragma unroll
+      for (int q = 0; q < 4; ++q) sn[q] = q < nb ? src[e + q] : sn[0];
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        const float* hv = h_src + (sn[q] * H + h) * C;
+        hh0[q] = lane < C ? hv[lane] : 0.f;
+        hh1[q] = kWave + lane < C ? hv[kWave + lane] : 0.f;
       }
-      Z = Z * scale + p;
-      acc0 = acc0 * scale + p * h0;
-      acc1 = acc1 * scale + p * h1;
+      float sc4[4];
+#pragma unroll
+      for (int q = 0; q < 4; ++q) sc4[q] = hh0[q] * as0 + hh1[q] * as1;
+#pragma unroll
+      for (int sft = kWave / 2; sft > 0; sft >>= 1) {
+#pragma unroll
+        for (int q = 0; q < 4; ++q)
+          sc4[q] += __shfl_down(sc4[q], sft);
+      }
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        if (q >= nb) continue;
+        float sv = ad + __shfl(sc4[q], 0);
+        if (lane == 0) spre_out[(e + q) * H + h] = sv;
+        sv = sv > 0.f ? sv : sv * slope;
+        float scale = 1.f;
+        float p;
+        if (sv > m) {
+          scale = __expf(m - sv);
+          p = 1.f;
+          m = sv;
+        } else {
+          p = __expf(sv - m);
+        }
+        Z = Z * scale + p;
+        acc0 = acc0 * scale + p * hh0[q];
+        acc1 = acc1 * scale + p * hh1[q];
+      }
     }
     const float inv = Z > 0.f ? 1.f / Z : 0.f;
     float* ov = out + (t * H + h) * C;
@@ -141,25 +163,53 @@ __global__ void gat_fused_bwd_kernel(
         (kWave + lane < C ? d1 * ov[kWave + lane] : 0.f));
     float dad_acc = 0.f;        // sum of ds over the segment
     float das0 = 0.f, das1 = 0.f;  // datt_src accumulator (per lane)
-    for (int64_t e = s0; e < s1; ++e) {
-      const int64_t sn = src[e];
-      const float* hv = h_src + (sn * H + h) * C;
-      const float h0 = lane < C ? hv[lane] : 0.f;
-      const float h1 = kWave + lane < C ? hv[kWave + lane] : 0.f;
-      // logits were cached by the forward pass: no recompute chain here
-      const float s_pre = spre_in[e * H + h];
-      const float s = s_pre > 0.f ? s_pre : s_pre * slope;
-      const float p = __expf(s - m) * inv;
-      const float dot_h = wave_sum(h0 * d0 + h1 * d1);
-      float ds = p * (dot_h - dot_o);
-      ds *= (s_pre > 0.f ? 1.f : slope);
-      float* dhv = dh_src + (sn * H + h) * C;
-      if (lane < C) atomicAdd(&dhv[lane], p * d0 + ds * as0);
-      if (kWave + lane < C)
-        atomicAdd(&dhv[kWave + lane], p * d1 + ds * as1);
-      das0 += ds * h0;
-      das1 += ds * h1;
-      dad_acc += ds;
+    // 4-edge software pipeline: the segment loop was a serial chain of
+    // dependent global loads (src[e] -> h_src row -> math), ~10 L2-miss
+    // latencies back to back per wave; batching the loads of 4 edges
+    // gives the memory system 4 independent misses in flight and lets
+    // the 4 wave-reductions interleave.
+    for (int64_t e = s0; e < s1; e += 4) {
+      const int nb = (int)((s1 - e) < 4 ? (s1 - e) : 4);
+      int64_t sn[4];
+      float hh0[4], hh1[4], spre4[4];
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        sn[q] = q < nb ? src[e + q] : sn[0];
+        if (q < nb) spre4[q] = spre_in[(e + q) * H + h];
+      }
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        const float* hv = h_src + (sn[q] * H + h) * C;
+        hh0[q] = lane < C ? hv[lane] : 0.f;
+        hh1[q] = kWave + lane < C ? hv[kWave + lane] : 0.f;
+      }
+      float dot_h4[4];
+#pragma unroll
+      for (int q = 0; q < 4; ++q)
+        dot_h4[q] = hh0[q] * d0 + hh1[q] * d1;
+      // interleaved wave reductions (independent shfl chains)
+#pragma unroll
+      for (int sft = kWave / 2; sft > 0; sft >>= 1) {
+#pragma unroll
+        for (int q = 0; q < 4; ++q)
+          dot_h4[q] += __shfl_down(dot_h4[q], sft);
+      }
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        if (q >= nb) continue;
+        const float s_pre = spre4[q];
+        const float sa = s_pre > 0.f ? s_pre : s_pre * slope;
+        const float p = __expf(sa - m) * inv;
+        float ds = p * (__shfl(dot_h4[q], 0) - dot_o);
+        ds *= (s_pre > 0.f ? 1.f : slope);
+        float* dhv = dh_src + (sn[q] * H + h) * C;
+        if (lane < C) atomicAdd(&dhv[lane], p * d0 + ds * as0);
+        if (kWave + lane < C)
+          atomicAdd(&dhv[kWave + lane], p * d1 + ds * as1);
+        das0 += ds * hh0[q];
+        das1 += ds * hh1[q];
+        dad_acc += ds;
+      }
     }
     // one atomic flush per wave (not per edge)
     float* dtv = dh_tgt + (t * H + h) * C;
