@@ -129,18 +129,29 @@ __global__ void gat_fused_bwd_kernel(
     const float* __restrict__ out, const float* __restrict__ m_in,
     const float* __restrict__ z_in, const float* __restrict__ spre_in,
     const float* __restrict__ dout,
-    int64_t n_tgt, int64_t H, int64_t C, float slope,
+    int64_t n_tgt, int64_t H, int64_t C, float slope, int64_t S,
     float* __restrict__ dh_tgt, float* __restrict__ dh_src,
     float* __restrict__ datt_src, float* __restrict__ datt_dst) {
   const int lane = threadIdx.x & (kWave - 1);
   const int64_t wave =
       (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) / kWave;
   const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) / kWave;
-  const int64_t total = n_tgt * H;
+  // S-way segment split: small-target layers (hop 0: 512 targets x 4
+  // heads = 2048 waves) cannot fill 256 CUs with (t, h) waves alone —
+  // every per-edge term in the backward is independent given the saved
+  // (m, Z, out) statistics, so S sub-waves each take a slice of the
+  // segment and flush their partial reductions atomically.
+  const int64_t total = n_tgt * H * S;
   for (int64_t w = wave; w < total; w += n_waves) {
-    const int64_t t = w / H;
-    const int64_t h = w - t * H;
-    const int64_t s0 = offsets[t], s1 = offsets[t + 1];
+    const int64_t t = w / (H * S);
+    const int64_t rem = w - t * H * S;
+    const int64_t h = rem / S;
+    const int64_t q = rem - h * S;
+    const int64_t f0 = offsets[t], f1 = offsets[t + 1];
+    if (f1 <= f0) continue;
+    const int64_t per = (f1 - f0 + S - 1) / S;
+    const int64_t s0 = f0 + q * per;
+    const int64_t s1 = s0 + per < f1 ? s0 + per : f1;
     if (s1 <= s0) continue;
     const float* ats = att_src + h * C;
     const float* atd = att_dst + h * C;
@@ -277,7 +288,11 @@ hip_gat_fused_bwd(const torch::Tensor& h_tgt, const torch::Tensor& h_src,
   auto das = torch::zeros_like(att_src);
   auto dad = torch::zeros_like(att_dst);
   if (n_tgt > 0) {
-    hipLaunchKernelGGL(gat_fused_bwd_kernel, dim3(gat_grid(n_tgt * H)),
+    // fill the chip: split segments when (t, h) waves alone are few
+    int64_t S = 32768 / std::max<int64_t>(n_tgt * H, 1);
+    S = std::max<int64_t>(1, std::min<int64_t>(S, 8));
+    hipLaunchKernelGGL(gat_fused_bwd_kernel,
+                       dim3(gat_grid(n_tgt * H * S)),
                        dim3(kBlock), 0, current_stream(),
                        h_tgt.data_ptr<float>(), h_src.data_ptr<float>(),
                        att_src.data_ptr<float>(),
@@ -286,7 +301,7 @@ hip_gat_fused_bwd(const torch::Tensor& h_tgt, const torch::Tensor& h_src,
                        out.data_ptr<float>(), m.data_ptr<float>(),
                        z.data_ptr<float>(), spre.data_ptr<float>(),
                        dout.contiguous().data_ptr<float>(), n_tgt, H, C,
-                       (float)slope, dh_tgt.data_ptr<float>(),
+                       (float)slope, S, dh_tgt.data_ptr<float>(),
                        dh_src.data_ptr<float>(), das.data_ptr<float>(),
                        dad.data_ptr<float>());
   }

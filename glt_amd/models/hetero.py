@@ -58,7 +58,10 @@ class HeteroConv(nn.Module):
                     self_out[t] = self_lins[t](x)
         result = {}
         for t, hs in out.items():
-            result[t] = torch.stack(hs).sum(0) if len(hs) > 1 else hs[0]
+            acc = hs[0]
+            for v in hs[1:]:  # chained adds: no stack copy, same grads
+                acc = acc + v
+            result[t] = acc
         return result if self_out is None else (result, self_out)
 
     def _batched_gat(self, x_dict, rels, self_lins=None, self_out=None):
@@ -92,15 +95,20 @@ class HeteroConv(nn.Module):
             W = weights[0] if len(weights) == 1 \
                 else torch.cat(weights, dim=0)
             h = torch.nn.functional.linear(x, W)
-            off = 0
-            for k, c in zip(keys, convs):
-                hc = c.heads * c.out_channels
-                # last-dim split of a stride-1 slice: a true view, no copy
-                H[(t, k)] = h[:, off:off + hc].view(
-                    x.size(0), c.heads, c.out_channels)
-                off += hc
+            # torch.split: the backward is ONE cat of the slice grads —
+            # manual h[:, a:b] views made autograd materialize zeros(h)
+            # + add per relation (a large share of RGAT's 49 fills and
+            # 20 adds per step; see profiles/r02 RGAT notes)
+            sizes = [c.heads * c.out_channels for c in convs]
             if fold_self:
-                s = h[:, off:]
+                sizes.append(h.size(1) - sum(sizes))
+            parts = torch.split(h, sizes, dim=1) if len(sizes) > 1 \
+                else (h,)
+            for k, c, hp in zip(keys, convs, parts):
+                H[(t, k)] = hp.contiguous().view(
+                    x.size(0), c.heads, c.out_channels)
+            if fold_self:
+                s = parts[-1]
                 b = self_lins[t].bias
                 self_out[t] = s if b is None else s + b
         out: Dict[NodeType, List[torch.Tensor]] = {}
