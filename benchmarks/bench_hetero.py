@@ -24,6 +24,10 @@ def main():
                     choices=["rgat", "rsage"])
     ap.add_argument("--fanout", type=str, default="10,5")
     ap.add_argument("--batch-size", type=int, default=512)
+    ap.add_argument("--dtype", type=str, default="fp32",
+                    choices=["fp32", "bf16"],
+                    help="bf16: features stored bf16, bf16 GAT/segment "
+                         "kernels + cast-linear over fp32 master params")
     ap.add_argument("--steps", type=int, default=60)
     ap.add_argument("--warmup", type=int, default=10)
     ap.add_argument("--prefetch", type=int, default=3)
@@ -64,11 +68,12 @@ def main():
                              "institute": n_inst, "fos": n_fos},
                   device=0 if has_gpu else None)
     dim = args.feat_dim
+    fdt = torch.bfloat16 if args.dtype == "bf16" else torch.float32
     ds.init_node_features(
-        {"paper": torch.randn(n_paper, dim),
-         "author": torch.randn(n_author, dim),
-         "institute": torch.randn(n_inst, dim),
-         "fos": torch.randn(n_fos, dim)},
+        {"paper": torch.randn(n_paper, dim).to(fdt),
+         "author": torch.randn(n_author, dim).to(fdt),
+         "institute": torch.randn(n_inst, dim).to(fdt),
+         "fos": torch.randn(n_fos, dim).to(fdt)},
         split_ratio=1.0 if has_gpu else 0.0,
         device=0 if has_gpu else None, with_gpu=has_gpu)
     ds.init_node_labels({"paper": torch.randint(0, args.classes,
@@ -106,7 +111,7 @@ def main():
         out = model(data.x_dict, data.edge_index_dict,
                     predict_type="paper")
         bs = data["paper"].batch_size
-        loss = F.cross_entropy(out[:bs], data["paper"].y[:bs])
+        loss = F.cross_entropy(out[:bs].float(), data["paper"].y[:bs])
         loss.backward()
         opt.step()
 
@@ -124,7 +129,7 @@ def main():
         "metric": f"{args.model} IGBH-shaped hetero train batches/sec",
         "value": round(args.steps / dt, 3),
         "ms_per_step": round(dt / args.steps * 1e3, 3),
-        "config": {"papers": n_paper, "feat_dim": dim,
+        "config": {"dtype": args.dtype, "papers": n_paper, "feat_dim": dim,
                    "fanout": fanout, "batch": args.batch_size,
                    "model": args.model, "heads": args.heads},
     }))

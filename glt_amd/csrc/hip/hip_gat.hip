@@ -24,6 +24,8 @@
  *   datt_src[h]    += ds_e * h_src[src_e];  dh_tgt[t] += (sum ds) * att_dst
  *   datt_dst[h]    += (sum_e ds_e) * h_tgt[t]
  */
+#include <hip/hip_bf16.h>
+
 #include "hip_common.h"
 #include "../include/common.h"
 
@@ -37,15 +39,16 @@ __device__ __forceinline__ float wave_sum(float v) {
   return __shfl(v, 0);
 }
 
+template <typename T>  // h/out dtype: float or bf16 (math stays fp32)
 __global__ void gat_fused_fwd_kernel(
-    const float* __restrict__ h_tgt,      // [Nt.., H, C]
-    const float* __restrict__ h_src,      // [Ns, H, C]
+    const T* __restrict__ h_tgt,          // [Nt.., H, C]
+    const T* __restrict__ h_src,          // [Ns, H, C]
     const float* __restrict__ att_src,    // [H, C]
     const float* __restrict__ att_dst,    // [H, C]
     const int64_t* __restrict__ src,      // [E]
     const int64_t* __restrict__ offsets,  // [Nt+1]
     int64_t n_tgt, int64_t H, int64_t C, float slope,
-    float* __restrict__ out,              // [Nt, H, C]
+    T* __restrict__ out,                  // [Nt, H, C]
     float* __restrict__ m_out,            // [Nt, H]
     float* __restrict__ z_out,            // [Nt, H]
     float* __restrict__ spre_out) {       // [E, H] pre-activation logits
@@ -62,10 +65,11 @@ __global__ void gat_fused_fwd_kernel(
     const float* atd = att_dst + h * C;
     const float as0 = lane < C ? ats[lane] : 0.f;
     const float as1 = kWave + lane < C ? ats[kWave + lane] : 0.f;
-    const float* tv = h_tgt + (t * H + h) * C;
+    const T* tv = h_tgt + (t * H + h) * C;
     const float ad = wave_sum(
-        (lane < C ? tv[lane] * atd[lane] : 0.f) +
-        (kWave + lane < C ? tv[kWave + lane] * atd[kWave + lane] : 0.f));
+        (lane < C ? (float)tv[lane] * atd[lane] : 0.f) +
+        (kWave + lane < C ? (float)tv[kWave + lane] * atd[kWave + lane]
+                          : 0.f));
     float m = -1e30f, Z = 0.f;
     float acc0 = 0.f, acc1 = 0.f;  // lanes cover C (up to 2 passes)
     // 4-edge load batching (see backward): the online-softmax chain
@@ -78,9 +82,9 @@ __global__ void gat_fused_fwd_kernel(
       for (int q = 0; q < 4; ++q) sn[q] = q < nb ? src[e + q] : sn[0];
 #pragma unroll
       for (int q = 0; q < 4; ++q) {
-        const float* hv = h_src + (sn[q] * H + h) * C;
-        hh0[q] = lane < C ? hv[lane] : 0.f;
-        hh1[q] = kWave + lane < C ? hv[kWave + lane] : 0.f;
+        const T* hv = h_src + (sn[q] * H + h) * C;
+        hh0[q] = lane < C ? (float)hv[lane] : 0.f;
+        hh1[q] = kWave + lane < C ? (float)hv[kWave + lane] : 0.f;
       }
       float sc4[4];
 #pragma unroll
@@ -112,9 +116,9 @@ __global__ void gat_fused_fwd_kernel(
       }
     }
     const float inv = Z > 0.f ? 1.f / Z : 0.f;
-    float* ov = out + (t * H + h) * C;
-    if (lane < C) ov[lane] = acc0 * inv;
-    if (kWave + lane < C) ov[kWave + lane] = acc1 * inv;
+    T* ov = out + (t * H + h) * C;
+    if (lane < C) ov[lane] = (T)(acc0 * inv);
+    if (kWave + lane < C) ov[kWave + lane] = (T)(acc1 * inv);
     if (lane == 0) {
       m_out[t * H + h] = m;
       z_out[t * H + h] = Z;
@@ -122,13 +126,14 @@ __global__ void gat_fused_fwd_kernel(
   }
 }
 
+template <typename T>
 __global__ void gat_fused_bwd_kernel(
-    const float* __restrict__ h_tgt, const float* __restrict__ h_src,
+    const T* __restrict__ h_tgt, const T* __restrict__ h_src,
     const float* __restrict__ att_src, const float* __restrict__ att_dst,
     const int64_t* __restrict__ src, const int64_t* __restrict__ offsets,
-    const float* __restrict__ out, const float* __restrict__ m_in,
+    const T* __restrict__ out, const float* __restrict__ m_in,
     const float* __restrict__ z_in, const float* __restrict__ spre_in,
-    const float* __restrict__ dout,
+    const T* __restrict__ dout,
     int64_t n_tgt, int64_t H, int64_t C, float slope, int64_t S,
     float* __restrict__ dh_tgt, float* __restrict__ dh_src,
     float* __restrict__ datt_src, float* __restrict__ datt_dst) {
@@ -157,21 +162,21 @@ __global__ void gat_fused_bwd_kernel(
     const float* atd = att_dst + h * C;
     const float as0 = lane < C ? ats[lane] : 0.f;
     const float as1 = kWave + lane < C ? ats[kWave + lane] : 0.f;
-    const float* tv = h_tgt + (t * H + h) * C;
-    const float t0 = lane < C ? tv[lane] : 0.f;
-    const float t1 = kWave + lane < C ? tv[kWave + lane] : 0.f;
+    const T* tv = h_tgt + (t * H + h) * C;
+    const float t0 = lane < C ? (float)tv[lane] : 0.f;
+    const float t1 = kWave + lane < C ? (float)tv[kWave + lane] : 0.f;
     const float ad0 = lane < C ? atd[lane] : 0.f;
     const float ad1 = kWave + lane < C ? atd[kWave + lane] : 0.f;
     const float m = m_in[t * H + h];
     const float Z = z_in[t * H + h];
     const float inv = Z > 0.f ? 1.f / Z : 0.f;
-    const float* dv = dout + (t * H + h) * C;
-    const float* ov = out + (t * H + h) * C;
-    const float d0 = lane < C ? dv[lane] : 0.f;
-    const float d1 = kWave + lane < C ? dv[kWave + lane] : 0.f;
+    const T* dv = dout + (t * H + h) * C;
+    const T* ov = out + (t * H + h) * C;
+    const float d0 = lane < C ? (float)dv[lane] : 0.f;
+    const float d1 = kWave + lane < C ? (float)dv[kWave + lane] : 0.f;
     const float dot_o = wave_sum(
-        (lane < C ? d0 * ov[lane] : 0.f) +
-        (kWave + lane < C ? d1 * ov[kWave + lane] : 0.f));
+        (lane < C ? d0 * (float)ov[lane] : 0.f) +
+        (kWave + lane < C ? d1 * (float)ov[kWave + lane] : 0.f));
     float dad_acc = 0.f;        // sum of ds over the segment
     float das0 = 0.f, das1 = 0.f;  // datt_src accumulator (per lane)
     // 4-edge software pipeline: the segment loop was a serial chain of
@@ -190,9 +195,9 @@ __global__ void gat_fused_bwd_kernel(
       }
 #pragma unroll
       for (int q = 0; q < 4; ++q) {
-        const float* hv = h_src + (sn[q] * H + h) * C;
-        hh0[q] = lane < C ? hv[lane] : 0.f;
-        hh1[q] = kWave + lane < C ? hv[kWave + lane] : 0.f;
+        const T* hv = h_src + (sn[q] * H + h) * C;
+        hh0[q] = lane < C ? (float)hv[lane] : 0.f;
+        hh1[q] = kWave + lane < C ? (float)hv[kWave + lane] : 0.f;
       }
       float dot_h4[4];
 #pragma unroll
@@ -255,20 +260,39 @@ hip_gat_fused_fwd(
   const int64_t E = src.numel();
   TORCH_CHECK(C <= 2 * kWave, "GAT fused kernel supports C <= 128");
   TORCH_CHECK(h_tgt.size(0) >= n_tgt, "h_tgt must cover all targets");
+  const bool bf16 = h_src.scalar_type() == torch::kBFloat16;
+  auto fopt = h_src.options().dtype(torch::kFloat32);
   auto out = torch::empty({n_tgt, H, C}, h_src.options());
-  auto m = torch::empty({n_tgt, H}, h_src.options());
-  auto z = torch::empty({n_tgt, H}, h_src.options());
-  auto spre = torch::empty({E, H}, h_src.options());
+  auto m = torch::empty({n_tgt, H}, fopt);
+  auto z = torch::empty({n_tgt, H}, fopt);
+  auto spre = torch::empty({E, H}, fopt);
+  auto as = att_src.to(torch::kFloat32).contiguous();
+  auto ad = att_dst.to(torch::kFloat32).contiguous();
   if (n_tgt > 0) {
-    hipLaunchKernelGGL(gat_fused_fwd_kernel, dim3(gat_grid(n_tgt * H)),
-                       dim3(kBlock), 0, current_stream(),
-                       h_tgt.data_ptr<float>(), h_src.data_ptr<float>(),
-                       att_src.data_ptr<float>(),
-                       att_dst.data_ptr<float>(), src.data_ptr<int64_t>(),
-                       offsets.data_ptr<int64_t>(), n_tgt, H, C,
-                       (float)slope, out.data_ptr<float>(),
-                       m.data_ptr<float>(), z.data_ptr<float>(),
-                       spre.data_ptr<float>());
+    if (bf16) {
+      hipLaunchKernelGGL(gat_fused_fwd_kernel<__bf16>,
+                         dim3(gat_grid(n_tgt * H)),
+                         dim3(kBlock), 0, current_stream(),
+                         (const __bf16*)h_tgt.data_ptr(),
+                         (const __bf16*)h_src.data_ptr(),
+                         as.data_ptr<float>(), ad.data_ptr<float>(),
+                         src.data_ptr<int64_t>(),
+                         offsets.data_ptr<int64_t>(), n_tgt, H, C,
+                         (float)slope, (__bf16*)out.data_ptr(),
+                         m.data_ptr<float>(), z.data_ptr<float>(),
+                         spre.data_ptr<float>());
+    } else {
+      hipLaunchKernelGGL(gat_fused_fwd_kernel<float>,
+                         dim3(gat_grid(n_tgt * H)),
+                         dim3(kBlock), 0, current_stream(),
+                         h_tgt.data_ptr<float>(), h_src.data_ptr<float>(),
+                         as.data_ptr<float>(), ad.data_ptr<float>(),
+                         src.data_ptr<int64_t>(),
+                         offsets.data_ptr<int64_t>(), n_tgt, H, C,
+                         (float)slope, out.data_ptr<float>(),
+                         m.data_ptr<float>(), z.data_ptr<float>(),
+                         spre.data_ptr<float>());
+    }
   }
   return {out, m, z, spre};
 }
@@ -283,27 +307,52 @@ hip_gat_fused_bwd(const torch::Tensor& h_tgt, const torch::Tensor& h_src,
                   double slope) {
   const int64_t n_tgt = offsets.numel() - 1;
   const int64_t H = h_src.size(1), C = h_src.size(2);
-  auto dh_tgt = torch::zeros_like(h_tgt);
-  auto dh_src = torch::zeros_like(h_src);
-  auto das = torch::zeros_like(att_src);
-  auto dad = torch::zeros_like(att_dst);
+  const bool bf16 = h_src.scalar_type() == torch::kBFloat16;
+  auto fopt = h_src.options().dtype(torch::kFloat32);
+  // grad arenas are always fp32 (atomic accumulation); the python
+  // wrapper casts dh back to the h dtype once
+  auto dh_tgt = torch::zeros(h_tgt.sizes(), fopt);
+  auto dh_src = torch::zeros(h_src.sizes(), fopt);
+  auto das = torch::zeros(att_src.sizes(), fopt);
+  auto dad = torch::zeros(att_dst.sizes(), fopt);
+  auto as = att_src.to(torch::kFloat32).contiguous();
+  auto ad = att_dst.to(torch::kFloat32).contiguous();
   if (n_tgt > 0) {
     // fill the chip: split segments when (t, h) waves alone are few
     int64_t S = 32768 / std::max<int64_t>(n_tgt * H, 1);
     S = std::max<int64_t>(1, std::min<int64_t>(S, 8));
-    hipLaunchKernelGGL(gat_fused_bwd_kernel,
-                       dim3(gat_grid(n_tgt * H * S)),
-                       dim3(kBlock), 0, current_stream(),
-                       h_tgt.data_ptr<float>(), h_src.data_ptr<float>(),
-                       att_src.data_ptr<float>(),
-                       att_dst.data_ptr<float>(), src.data_ptr<int64_t>(),
-                       offsets.data_ptr<int64_t>(),
-                       out.data_ptr<float>(), m.data_ptr<float>(),
-                       z.data_ptr<float>(), spre.data_ptr<float>(),
-                       dout.contiguous().data_ptr<float>(), n_tgt, H, C,
-                       (float)slope, S, dh_tgt.data_ptr<float>(),
-                       dh_src.data_ptr<float>(), das.data_ptr<float>(),
-                       dad.data_ptr<float>());
+    auto dc = dout.contiguous();
+    if (bf16) {
+      hipLaunchKernelGGL(gat_fused_bwd_kernel<__bf16>,
+                         dim3(gat_grid(n_tgt * H * S)),
+                         dim3(kBlock), 0, current_stream(),
+                         (const __bf16*)h_tgt.data_ptr(),
+                         (const __bf16*)h_src.data_ptr(),
+                         as.data_ptr<float>(), ad.data_ptr<float>(),
+                         src.data_ptr<int64_t>(),
+                         offsets.data_ptr<int64_t>(),
+                         (const __bf16*)out.data_ptr(),
+                         m.data_ptr<float>(),
+                         z.data_ptr<float>(), spre.data_ptr<float>(),
+                         (const __bf16*)dc.data_ptr(), n_tgt, H, C,
+                         (float)slope, S, dh_tgt.data_ptr<float>(),
+                         dh_src.data_ptr<float>(), das.data_ptr<float>(),
+                         dad.data_ptr<float>());
+    } else {
+      hipLaunchKernelGGL(gat_fused_bwd_kernel<float>,
+                         dim3(gat_grid(n_tgt * H * S)),
+                         dim3(kBlock), 0, current_stream(),
+                         h_tgt.data_ptr<float>(), h_src.data_ptr<float>(),
+                         as.data_ptr<float>(), ad.data_ptr<float>(),
+                         src.data_ptr<int64_t>(),
+                         offsets.data_ptr<int64_t>(),
+                         out.data_ptr<float>(), m.data_ptr<float>(),
+                         z.data_ptr<float>(), spre.data_ptr<float>(),
+                         dc.data_ptr<float>(), n_tgt, H, C,
+                         (float)slope, S, dh_tgt.data_ptr<float>(),
+                         dh_src.data_ptr<float>(), das.data_ptr<float>(),
+                         dad.data_ptr<float>());
+    }
   }
   return {dh_tgt, dh_src, das, dad};
 }

@@ -55,7 +55,13 @@ class HeteroConv(nn.Module):
         if self_out is not None:
             for t, x in x_dict.items():
                 if t not in self_out and t in self_lins:
-                    self_out[t] = self_lins[t](x)
+                    lin = self_lins[t]
+                    if x.dtype != lin.weight.dtype:
+                        from ..ops import cast_linear
+
+                        self_out[t] = cast_linear(x, lin.weight, lin.bias)
+                    else:
+                        self_out[t] = lin(x)
         result = {}
         for t, hs in out.items():
             acc = hs[0]
@@ -84,8 +90,14 @@ class HeteroConv(nn.Module):
             in_dims = {c.lin.weight.size(1) for c in convs}
             if len(in_dims) > 1:
                 for k, c in zip(keys, convs):
-                    H[(t, k)] = c.lin(x).view(x.size(0), c.heads,
-                                              c.out_channels)
+                    if x.dtype != c.lin.weight.dtype:
+                        from ..ops import cast_linear
+
+                        hx = cast_linear(x, c.lin.weight, c.lin.bias)
+                    else:
+                        hx = c.lin(x)
+                    H[(t, k)] = hx.view(x.size(0), c.heads,
+                                        c.out_channels)
                 continue
             weights = [c.lin.weight for c in convs]
             fold_self = (self_lins is not None and t in self_lins
@@ -94,7 +106,12 @@ class HeteroConv(nn.Module):
                 weights.append(self_lins[t].weight)
             W = weights[0] if len(weights) == 1 \
                 else torch.cat(weights, dim=0)
-            h = torch.nn.functional.linear(x, W)
+            if x.dtype != W.dtype:
+                from ..ops import cast_linear
+
+                h = cast_linear(x, W, None)
+            else:
+                h = torch.nn.functional.linear(x, W)
             # torch.split: the backward is ONE cat of the slice grads —
             # manual h[:, a:b] views made autograd materialize zeros(h)
             # + add per relation (a large share of RGAT's 49 fills and
@@ -110,7 +127,7 @@ class HeteroConv(nn.Module):
             if fold_self:
                 s = parts[-1]
                 b = self_lins[t].bias
-                self_out[t] = s if b is None else s + b
+                self_out[t] = s if b is None else s + b.to(s.dtype)
         out: Dict[NodeType, List[torch.Tensor]] = {}
         for etype, key, ei in rels:
             src_t, _, dst_t = etype
@@ -191,6 +208,13 @@ class RGNN(nn.Module):
                 h_next[t] = F.dropout(F.relu(out), p=self.dropout,
                                       training=self.training)
             h = h_next
+        def _head(v):
+            if v.dtype != self.head.weight.dtype:
+                from ..ops import cast_linear
+
+                return cast_linear(v, self.head.weight, self.head.bias)
+            return self.head(v)
+
         if predict_type is not None:
-            return self.head(h[predict_type])
-        return {t: self.head(v) for t, v in h.items()}
+            return _head(h[predict_type])
+        return {t: _head(v) for t, v in h.items()}

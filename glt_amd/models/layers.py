@@ -151,17 +151,24 @@ class GATConv(nn.Module):
         edge sets not sorted by edge_index[0] — the fused segment kernel
         requires ascending targets; the scatter-softmax fallback does
         not."""
+        def _proj(v):
+            if v.dtype != self.lin.weight.dtype:
+                from ..ops import cast_linear
+
+                return cast_linear(v, self.lin.weight, self.lin.bias)
+            return self.lin(v)
+
         if isinstance(x, tuple):
             x_tgt, x_src = x
             nt = num_target if num_target is not None else x_tgt.size(0)
-            h_tgt = self.lin(x_tgt).view(x_tgt.size(0), self.heads,
-                                         self.out_channels)
-            h_src = self.lin(x_src).view(x_src.size(0), self.heads,
-                                         self.out_channels)
+            h_tgt = _proj(x_tgt).view(x_tgt.size(0), self.heads,
+                                      self.out_channels)
+            h_src = _proj(x_src).view(x_src.size(0), self.heads,
+                                      self.out_channels)
         else:
             nt = num_target if num_target is not None else x.size(0)
-            h_tgt = h_src = self.lin(x).view(x.size(0), self.heads,
-                                             self.out_channels)
+            h_tgt = h_src = _proj(x).view(x.size(0), self.heads,
+                                          self.out_channels)
         return self.attend(h_tgt, h_src, edge_index, nt,
                            sorted_by_target=sorted_by_target)
 
@@ -174,7 +181,7 @@ class GATConv(nn.Module):
         h = h_src
         if (sorted_by_target and getattr(self, "use_fused", True)
                 and h_src.is_cuda
-                and h_src.dtype == torch.float32
+                and h_src.dtype in (torch.float32, torch.bfloat16)
                 and self.out_channels <= 128
                 and not (self.training and self.dropout > 0)):
             # fused segment-softmax-aggregate (edges sorted by target);
@@ -190,8 +197,11 @@ class GATConv(nn.Module):
             out = out.reshape(nt, self.heads * self.out_channels) \
                 if self.concat else out.mean(dim=1)
             if self.bias is not None:
-                out = out + self.bias
+                out = out + self.bias.to(out.dtype)
             return out
+        if h_src.dtype != torch.float32:
+            h_tgt = h_tgt.float()
+            h_src = h = h_src.float()
         alpha_src = (h_src * self.att_src).sum(-1)
         alpha_dst = (h_tgt * self.att_dst).sum(-1)
         # index_select (not advanced indexing): its backward is an

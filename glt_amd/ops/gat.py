@@ -33,6 +33,13 @@ class _GatFused(torch.autograd.Function):
         dht, dhs, das, dad = _C.gat_fused_bwd(h_tgt, h_src, att_src,
                                               att_dst, src, offsets, out,
                                               m, z, spre, dout, ctx.slope)
+        # kernel accumulates in fp32 arenas; match autograd dtypes
+        if dht.dtype != h_tgt.dtype:
+            dht = dht.to(h_tgt.dtype)
+            dhs = dhs.to(h_src.dtype)
+        if das.dtype != att_src.dtype:
+            das = das.to(att_src.dtype)
+            dad = dad.to(att_dst.dtype)
         return dht, dhs, das, dad, None, None, None
 
 

@@ -924,3 +924,32 @@ def test_weighted_sampler_no_replace_gpu():
     # analytic expectation ~0.84 (matches the CPU E-S twin); uniform
     # draws would include any given id at ~1.6%
     assert 0.75 < heavy_rate < 0.95, heavy_rate
+
+
+def test_gat_fused_bf16_matches_fp32():
+    """bf16 GAT fused kernels (fp32 math/stats, fp32 grad arenas) vs the
+    fp32 fused path at bf16 rounding tolerance."""
+    from glt_amd.ops import gat_softmax_aggregate
+
+    torch.manual_seed(5)
+    n_src, n_tgt, E, Hh, C = 400, 64, 1200, 4, 64
+    h32 = torch.randn(n_src, Hh, C, device="cuda") * 0.3
+    a_s = torch.randn(Hh, C, device="cuda") * 0.2
+    a_d = torch.randn(Hh, C, device="cuda") * 0.2
+    tgt = torch.sort(torch.randint(0, n_tgt, (E,), device="cuda")).values
+    src = torch.randint(0, n_src, (E,), device="cuda")
+
+    hA = h32.clone().requires_grad_(True)
+    out32 = gat_softmax_aggregate(hA, hA, a_s, a_d, tgt, src, n_tgt, 0.2)
+    g = torch.randn_like(out32)
+    out32.backward(g)
+
+    h16 = h32.to(torch.bfloat16).requires_grad_(True)
+    out16 = gat_softmax_aggregate(h16, h16, a_s, a_d, tgt, src, n_tgt,
+                                  0.2)
+    assert out16.dtype == torch.bfloat16
+    assert (out16.float() - out32).abs().max() < 0.05
+    out16.backward(g.to(torch.bfloat16))
+    assert h16.grad.dtype == torch.bfloat16
+    scale = max(hA.grad.abs().max().item(), 1.0)
+    assert (h16.grad.float() - hA.grad).abs().max() / scale < 0.08
