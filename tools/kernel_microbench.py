@@ -41,6 +41,21 @@ def build_flagship_shapes(op, device):
         A = torch.randn(130_000, 512, device=device)
         B = torch.randn(128, 512, device=device)
         return (A, B)
+    if op == "gemm_bt_bf16":
+        A = torch.randn(164_000, 200, device=device).to(torch.bfloat16)
+        W = torch.randn(256, 200, device=device).to(torch.bfloat16)
+        b = torch.randn(256, device=device)
+        return (A, W, b)
+    if op == "gemm_kt_bf16":
+        A = torch.randn(164_000, 256, device=device).to(torch.bfloat16)
+        B = torch.randn(164_000, 200, device=device).to(torch.bfloat16)
+        return (A, B)
+    if op == "seg_mean_cat_bf16":
+        n_src, n_tgt, E, F = 590_000, 130_000, 890_000, 100
+        x = torch.randn(n_src, F, device=device).to(torch.bfloat16)
+        tgt = torch.sort(torch.randint(0, n_tgt, (E,), device=device))[0]
+        src = torch.randint(0, n_src, (E,), device=device)
+        return (x, tgt, src, n_tgt)
     if op == "gather":
         feats = torch.randn(2_449_029, 100, device=device)
         rows = torch.randint(0, feats.size(0), (736_000,), device=device)
@@ -76,7 +91,8 @@ def main():
                     choices=["seg_mean", "seg_mean_cat", "gat_fused",
                              "mfma_gemm", "gather", "sample",
                              "sample_weighted", "sample_weighted_nr",
-                             "sample_weighted_hub"])
+                             "sample_weighted_hub", "gemm_bt_bf16",
+                             "gemm_kt_bf16", "seg_mean_cat_bf16"])
     ap.add_argument("--iters", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=3)
     args = ap.parse_args()
@@ -106,6 +122,17 @@ def main():
     elif args.op == "mfma_gemm":
         A, B = inp
         fn = lambda: _C.sage_gemm(A, B, None, False)
+    elif args.op == "gemm_bt_bf16":
+        A, W, b = inp
+        fn = lambda: _C.gemm_bt_bf16(A, W, b, True, False)
+    elif args.op == "gemm_kt_bf16":
+        A, B = inp
+        fn = lambda: _C.gemm_kt_bf16(A, B, True)
+    elif args.op == "seg_mean_cat_bf16":
+        from glt_amd.ops.segment import _boundaries
+        x, tgt, src, n_tgt = inp
+        off = torch.searchsorted(tgt, _boundaries(n_tgt, device))
+        fn = lambda: _C.segment_mean_cat_fwd(x, src, off, n_tgt)
     elif args.op == "gather":
         feats, rows = inp
         store = _C.UnifiedFeatureStore(0)
