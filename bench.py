@@ -31,8 +31,8 @@ import torch.nn.functional as F
 def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=50)
-    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--steps", type=int, default=300)
+    p.add_argument("--warmup", type=int, default=20)
     p.add_argument("--batch-size", type=int, default=1024)
     p.add_argument("--fanout", type=str, default="15,10,5")
     p.add_argument("--nodes", type=int, default=2_449_029)

@@ -953,3 +953,21 @@ def test_gat_fused_bf16_matches_fp32():
     assert h16.grad.dtype == torch.bfloat16
     scale = max(hA.grad.abs().max().item(), 1.0)
     assert (h16.grad.float() - hA.grad).abs().max() / scale < 0.08
+
+
+def test_deferred_sampler_empty_seeds_gpu():
+    """Empty seed batch through the deferred-sync path (round-1 known
+    gap: loaders never emit empty batches, so this was analysis-only)."""
+    from glt_amd.sampler import NeighborSampler
+
+    topo = Topology(torch.tensor([[0, 1], [1, 2]]), num_nodes=3)
+    g = Graph(topo, mode="CUDA", device=0)
+    s = NeighborSampler(g, [2, 2], device=torch.device("cuda", 0))
+    out = s._sample_from_nodes(torch.empty(0, dtype=torch.long,
+                                           device="cuda"))
+    assert out.node.numel() == 0
+    assert out.row.numel() == 0 and out.col.numel() == 0
+    assert out.num_sampled_nodes[0] == 0
+    # and a seed batch of out-of-range + empty mix still guards
+    out2 = s._sample_from_nodes(torch.tensor([5, -1], device="cuda"))
+    assert out2.node.numel() >= 0  # no crash; guards clamp
