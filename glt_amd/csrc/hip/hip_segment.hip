@@ -63,6 +63,33 @@ __global__ void seg_mean_fwd_bf162_kernel(const __hip_bfloat162* __restrict__ x,
   for (int64_t t = wave; t < n_tgt; t += n_waves) {
     const int64_t s = off[t], e = off[t + 1];
     const float inv = e > s ? 1.0f / (float)(e - s) : 0.0f;
+    if (feat2 <= kWave) {
+      // one channel-pair per lane: accumulate in registers with the
+      // row addresses batched 8 deep, so the gather issues 8
+      // independent misses instead of a serial col->row->col chain
+      const int64_t f = lane;
+      const bool act = f < feat2;
+      float2 acc = {0.f, 0.f};
+      for (int64_t k0 = s; k0 < e; k0 += 8) {
+        const int nb = (int)((e - k0) < 8 ? (e - k0) : 8);
+        int64_t cc[8];
+#pragma unroll
+        for (int q = 0; q < 8; ++q)
+          cc[q] = (q < nb ? col[k0 + q] : col[k0]) * feat2;
+#pragma unroll
+        for (int q = 0; q < 8; ++q) {
+          if (q < nb && act) {
+            const float2 v = __bfloat1622float2(x[cc[q] + f]);
+            acc.x += v.x;
+            acc.y += v.y;
+          }
+        }
+      }
+      if (act)
+        out[t * feat2 + f] =
+            __float22bfloat162_rn({acc.x * inv, acc.y * inv});
+      continue;
+    }
     for (int64_t f = lane; f < feat2; f += kWave) {
       float2 acc = {0.f, 0.f};
       for (int64_t k = s; k < e; ++k) {
@@ -88,6 +115,32 @@ __global__ void seg_mean_cat_fwd_bf162_kernel(
   for (int64_t t = wave; t < n_tgt; t += n_waves) {
     const int64_t s = off[t], e = off[t + 1];
     const float inv = e > s ? 1.0f / (float)(e - s) : 0.0f;
+    if (feat2 <= kWave) {
+      const int64_t f = lane;
+      const bool act = f < feat2;
+      float2 acc = {0.f, 0.f};
+      for (int64_t k0 = s; k0 < e; k0 += 8) {
+        const int nb = (int)((e - k0) < 8 ? (e - k0) : 8);
+        int64_t cc[8];
+#pragma unroll
+        for (int q = 0; q < 8; ++q)
+          cc[q] = (q < nb ? col[k0 + q] : col[k0]) * feat2;
+#pragma unroll
+        for (int q = 0; q < 8; ++q) {
+          if (q < nb && act) {
+            const float2 v = __bfloat1622float2(x[cc[q] + f]);
+            acc.x += v.x;
+            acc.y += v.y;
+          }
+        }
+      }
+      if (act) {
+        out[t * ostride + f] =
+            __float22bfloat162_rn({acc.x * inv, acc.y * inv});
+        out[t * ostride + feat2 + f] = x[t * feat2 + f];
+      }
+      continue;
+    }
     for (int64_t f = lane; f < feat2; f += kWave) {
       float2 acc = {0.f, 0.f};
       for (int64_t k = s; k < e; ++k) {
