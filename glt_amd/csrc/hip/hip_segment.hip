@@ -39,6 +39,29 @@ __global__ void seg_mean_fwd_kernel(const scalar_t* __restrict__ x,
   for (int64_t t = wave; t < n_tgt; t += n_waves) {
     const int64_t s = off[t], e = off[t + 1];
     const float inv = e > s ? 1.0f / (float)(e - s) : 0.0f;
+    if (feat <= 2 * kWave) {
+      // lane owns channels f and f+64: register accumulators + row
+      // addresses batched 8 deep (independent gather misses)
+      const int64_t f0 = lane, f1 = lane + kWave;
+      float a0 = 0.f, a1 = 0.f;
+      for (int64_t k0 = s; k0 < e; k0 += 8) {
+        const int nb = (int)((e - k0) < 8 ? (e - k0) : 8);
+        int64_t cc[8];
+#pragma unroll
+        for (int q = 0; q < 8; ++q)
+          cc[q] = (q < nb ? col[k0 + q] : col[k0]) * feat;
+#pragma unroll
+        for (int q = 0; q < 8; ++q) {
+          if (q < nb) {
+            if (f0 < feat) a0 += static_cast<float>(x[cc[q] + f0]);
+            if (f1 < feat) a1 += static_cast<float>(x[cc[q] + f1]);
+          }
+        }
+      }
+      if (f0 < feat) out[t * feat + f0] = static_cast<scalar_t>(a0 * inv);
+      if (f1 < feat) out[t * feat + f1] = static_cast<scalar_t>(a1 * inv);
+      continue;
+    }
     for (int64_t f = lane; f < feat; f += kWave) {
       float acc = 0.f;
       for (int64_t k = s; k < e; ++k)
@@ -199,6 +222,33 @@ __global__ void seg_mean_cat_fwd_kernel(const scalar_t* __restrict__ x,
   for (int64_t t = wave; t < n_tgt; t += n_waves) {
     const int64_t s = off[t], e = off[t + 1];
     const float inv = e > s ? 1.0f / (float)(e - s) : 0.0f;
+    if (feat <= 2 * kWave) {
+      const int64_t f0 = lane, f1 = lane + kWave;
+      float a0 = 0.f, a1 = 0.f;
+      for (int64_t k0 = s; k0 < e; k0 += 8) {
+        const int nb = (int)((e - k0) < 8 ? (e - k0) : 8);
+        int64_t cc[8];
+#pragma unroll
+        for (int q = 0; q < 8; ++q)
+          cc[q] = (q < nb ? col[k0 + q] : col[k0]) * feat;
+#pragma unroll
+        for (int q = 0; q < 8; ++q) {
+          if (q < nb) {
+            if (f0 < feat) a0 += static_cast<float>(x[cc[q] + f0]);
+            if (f1 < feat) a1 += static_cast<float>(x[cc[q] + f1]);
+          }
+        }
+      }
+      if (f0 < feat) {
+        out[t * ostride + f0] = static_cast<scalar_t>(a0 * inv);
+        out[t * ostride + feat + f0] = x[t * feat + f0];
+      }
+      if (f1 < feat) {
+        out[t * ostride + f1] = static_cast<scalar_t>(a1 * inv);
+        out[t * ostride + feat + f1] = x[t * feat + f1];
+      }
+      continue;
+    }
     for (int64_t f = lane; f < feat; f += kWave) {
       float acc = 0.f;
       for (int64_t k = s; k < e; ++k)
