@@ -222,6 +222,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_bf16_selftest", &hip_mfma_bf16_selftest);
   m.def("gat_fused_fwd", &hip_gat_fused_fwd);
   m.def("gat_fused_bwd", &hip_gat_fused_bwd);
+  m.def("gat_multi_fwd", &hip_gat_multi_fwd);
+  m.def("gat_multi_bwd", &hip_gat_multi_bwd);
   m.def("segment_mean_fwd", &hip_segment_mean_fwd);
   m.def("segment_mean_bwd", &hip_segment_mean_bwd);
   m.def("segment_mean_cat_fwd", &hip_segment_mean_cat_fwd);

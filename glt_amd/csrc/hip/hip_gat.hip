@@ -250,6 +250,227 @@ int gat_grid(int64_t waves_needed) {
 
 }  // namespace
 
+namespace {
+
+// ---------------------------------------------------------------------------
+// Multi-relation fused GAT: ONE launch per hetero layer.  RGAT's step is
+// launch-bound (~285 kernels/step measured); per-relation attention
+// kernels, their offsets glue, the .contiguous() slice copies and the
+// slice-grad zeros+adds all collapse into one fwd + one bwd call over a
+// packed relation table.  h rows are accessed STRIDED (base + n*stride
+// + h*C + c), so the per-type batched-GEMM output [N_t, R_t*H*C] is
+// consumed in place and dh accumulates atomically straight into one
+// per-type arena.
+// ---------------------------------------------------------------------------
+constexpr int kMaxRel = 8;
+
+template <typename T>
+struct GatPack {
+  const T* h_tgt[kMaxRel];
+  const T* h_src[kMaxRel];
+  int64_t tgt_stride[kMaxRel];
+  int64_t src_stride[kMaxRel];
+  const float* att_src[kMaxRel];
+  const float* att_dst[kMaxRel];
+  const int64_t* src[kMaxRel];
+  const int64_t* off[kMaxRel];
+  T* out[kMaxRel];              // contiguous [n_tgt_r, H, C] slices
+  float* m[kMaxRel];
+  float* z[kMaxRel];
+  float* spre[kMaxRel];
+  float* dh_tgt[kMaxRel];       // bwd only (fp32 arenas, strided)
+  float* dh_src[kMaxRel];
+  const T* dout[kMaxRel];       // bwd only (contiguous slices)
+  float* datt_src[kMaxRel];
+  float* datt_dst[kMaxRel];
+  int64_t n_tgt[kMaxRel];
+  int64_t cum[kMaxRel + 1];     // cumulative n_tgt (work-item decode)
+  int n_rel;
+};
+
+template <typename T>
+__global__ void gat_multi_fwd_kernel(GatPack<T> P, int64_t H, int64_t C,
+                                     float slope, int64_t total) {
+  const int lane = threadIdx.x & (kWave - 1);
+  const int64_t wave =
+      (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) / kWave;
+  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) / kWave;
+  for (int64_t w = wave; w < total * H; w += n_waves) {
+    const int64_t tg = w / H;
+    const int64_t h = w - tg * H;
+    int r = 0;
+    while (r + 1 < P.n_rel && tg >= P.cum[r + 1]) ++r;
+    const int64_t t = tg - P.cum[r];
+    const int64_t s0 = P.off[r][t], s1 = P.off[r][t + 1];
+    const float* ats = P.att_src[r] + h * C;
+    const float* atd = P.att_dst[r] + h * C;
+    const float as0 = lane < C ? ats[lane] : 0.f;
+    const float as1 = kWave + lane < C ? ats[kWave + lane] : 0.f;
+    const T* tv = P.h_tgt[r] + t * P.tgt_stride[r] + h * C;
+    const float ad = wave_sum(
+        (lane < C ? (float)tv[lane] * atd[lane] : 0.f) +
+        (kWave + lane < C ? (float)tv[kWave + lane] * atd[kWave + lane]
+                          : 0.f));
+    float m = -1e30f, Z = 0.f;
+    float acc0 = 0.f, acc1 = 0.f;
+    const int64_t hstride = P.src_stride[r];
+    const T* hbase = P.h_src[r] + h * C;
+    for (int64_t e = s0; e < s1; e += 4) {
+      const int nb = (int)((s1 - e) < 4 ? (s1 - e) : 4);
+      int64_t sn[4];
+      float hh0[4], hh1[4];
+#pragma unroll
+      for (int q = 0; q < 4; ++q)
+        sn[q] = q < nb ? P.src[r][e + q] : sn[0];
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        const T* hv = hbase + sn[q] * hstride;
+        hh0[q] = lane < C ? (float)hv[lane] : 0.f;
+        hh1[q] = kWave + lane < C ? (float)hv[kWave + lane] : 0.f;
+      }
+      float sc4[4];
+#pragma unroll
+      for (int q = 0; q < 4; ++q) sc4[q] = hh0[q] * as0 + hh1[q] * as1;
+#pragma unroll
+      for (int sft = kWave / 2; sft > 0; sft >>= 1) {
+#pragma unroll
+        for (int q = 0; q < 4; ++q)
+          sc4[q] += __shfl_down(sc4[q], sft);
+      }
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        if (q >= nb) continue;
+        float sv = ad + __shfl(sc4[q], 0);
+        if (lane == 0) P.spre[r][(e + q) * H + h] = sv;
+        sv = sv > 0.f ? sv : sv * slope;
+        float scale = 1.f, pp;
+        if (sv > m) {
+          scale = __expf(m - sv);
+          pp = 1.f;
+          m = sv;
+        } else {
+          pp = __expf(sv - m);
+        }
+        Z = Z * scale + pp;
+        acc0 = acc0 * scale + pp * hh0[q];
+        acc1 = acc1 * scale + pp * hh1[q];
+      }
+    }
+    const float inv = Z > 0.f ? 1.f / Z : 0.f;
+    T* ov = P.out[r] + (t * H + h) * C;
+    if (lane < C) ov[lane] = (T)(acc0 * inv);
+    if (kWave + lane < C) ov[kWave + lane] = (T)(acc1 * inv);
+    if (lane == 0) {
+      P.m[r][t * H + h] = m;
+      P.z[r][t * H + h] = Z;
+    }
+  }
+}
+
+template <typename T>
+__global__ void gat_multi_bwd_kernel(GatPack<T> P, int64_t H, int64_t C,
+                                     float slope, int64_t S,
+                                     int64_t total) {
+  const int lane = threadIdx.x & (kWave - 1);
+  const int64_t wave =
+      (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) / kWave;
+  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) / kWave;
+  for (int64_t w = wave; w < total * H * S; w += n_waves) {
+    const int64_t tg = w / (H * S);
+    const int64_t rem = w - tg * H * S;
+    const int64_t h = rem / S;
+    const int64_t q0 = rem - h * S;
+    int r = 0;
+    while (r + 1 < P.n_rel && tg >= P.cum[r + 1]) ++r;
+    const int64_t t = tg - P.cum[r];
+    const int64_t f0 = P.off[r][t], f1 = P.off[r][t + 1];
+    if (f1 <= f0) continue;
+    const int64_t per = (f1 - f0 + S - 1) / S;
+    const int64_t s0 = f0 + q0 * per;
+    const int64_t s1 = s0 + per < f1 ? s0 + per : f1;
+    if (s1 <= s0) continue;
+    const float* ats = P.att_src[r] + h * C;
+    const float* atd = P.att_dst[r] + h * C;
+    const float as0 = lane < C ? ats[lane] : 0.f;
+    const float as1 = kWave + lane < C ? ats[kWave + lane] : 0.f;
+    const T* tv = P.h_tgt[r] + t * P.tgt_stride[r] + h * C;
+    const float t0 = lane < C ? (float)tv[lane] : 0.f;
+    const float t1 = kWave + lane < C ? (float)tv[kWave + lane] : 0.f;
+    const float ad0 = lane < C ? atd[lane] : 0.f;
+    const float ad1 = kWave + lane < C ? atd[kWave + lane] : 0.f;
+    const float m = P.m[r][t * H + h];
+    const float Z = P.z[r][t * H + h];
+    const float inv = Z > 0.f ? 1.f / Z : 0.f;
+    const T* dv = P.dout[r] + (t * H + h) * C;
+    const T* ov = P.out[r] + (t * H + h) * C;
+    const float d0 = lane < C ? (float)dv[lane] : 0.f;
+    const float d1 = kWave + lane < C ? (float)dv[kWave + lane] : 0.f;
+    const float dot_o = wave_sum(
+        (lane < C ? d0 * (float)ov[lane] : 0.f) +
+        (kWave + lane < C ? d1 * (float)ov[kWave + lane] : 0.f));
+    float dad_acc = 0.f;
+    float das0 = 0.f, das1 = 0.f;
+    const int64_t hstride = P.src_stride[r];
+    const T* hbase = P.h_src[r] + h * C;
+    float* dhbase = P.dh_src[r] + h * C;
+    for (int64_t e = s0; e < s1; e += 4) {
+      const int nb = (int)((s1 - e) < 4 ? (s1 - e) : 4);
+      int64_t sn[4];
+      float hh0[4], hh1[4], spre4[4];
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        sn[q] = q < nb ? P.src[r][e + q] : sn[0];
+        if (q < nb) spre4[q] = P.spre[r][(e + q) * H + h];
+      }
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        const T* hv = hbase + sn[q] * hstride;
+        hh0[q] = lane < C ? (float)hv[lane] : 0.f;
+        hh1[q] = kWave + lane < C ? (float)hv[kWave + lane] : 0.f;
+      }
+      float dot_h4[4];
+#pragma unroll
+      for (int q = 0; q < 4; ++q)
+        dot_h4[q] = hh0[q] * d0 + hh1[q] * d1;
+#pragma unroll
+      for (int sft = kWave / 2; sft > 0; sft >>= 1) {
+#pragma unroll
+        for (int q = 0; q < 4; ++q)
+          dot_h4[q] += __shfl_down(dot_h4[q], sft);
+      }
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        if (q >= nb) continue;
+        const float s_pre = spre4[q];
+        const float sa = s_pre > 0.f ? s_pre : s_pre * slope;
+        const float pp = __expf(sa - m) * inv;
+        float ds = pp * (__shfl(dot_h4[q], 0) - dot_o);
+        ds *= (s_pre > 0.f ? 1.f : slope);
+        float* dhv = dhbase + sn[q] * hstride;
+        if (lane < C) atomicAdd(&dhv[lane], pp * d0 + ds * as0);
+        if (kWave + lane < C)
+          atomicAdd(&dhv[kWave + lane], pp * d1 + ds * as1);
+        das0 += ds * hh0[q];
+        das1 += ds * hh1[q];
+        dad_acc += ds;
+      }
+    }
+    float* dtv = P.dh_tgt[r] + t * P.tgt_stride[r] + h * C;
+    if (lane < C) {
+      atomicAdd(&dtv[lane], dad_acc * ad0);
+      atomicAdd(&P.datt_src[r][h * C + lane], das0);
+      atomicAdd(&P.datt_dst[r][h * C + lane], dad_acc * t0);
+    }
+    if (kWave + lane < C) {
+      atomicAdd(&dtv[kWave + lane], dad_acc * ad1);
+      atomicAdd(&P.datt_src[r][h * C + kWave + lane], das1);
+      atomicAdd(&P.datt_dst[r][h * C + kWave + lane], dad_acc * t1);
+    }
+  }
+}
+
+}  // namespace
+
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
 hip_gat_fused_fwd(
     const torch::Tensor& h_tgt, const torch::Tensor& h_src,
@@ -355,6 +576,145 @@ hip_gat_fused_bwd(const torch::Tensor& h_tgt, const torch::Tensor& h_src,
     }
   }
   return {dh_tgt, dh_src, das, dad};
+}
+
+
+// ---------------------------------------------------------------------------
+// Multi-relation hosts.  h views may be STRIDED row-wise (slices of the
+// per-type batched projection [N_t, R_t*H*C]); dims 1,2 must be dense.
+// ---------------------------------------------------------------------------
+namespace {
+
+template <typename T>
+void fill_pack_common(GatPack<T>& P,
+                      const std::vector<torch::Tensor>& h_tgt,
+                      const std::vector<torch::Tensor>& h_src,
+                      const std::vector<torch::Tensor>& att_src,
+                      const std::vector<torch::Tensor>& att_dst,
+                      const std::vector<torch::Tensor>& src,
+                      const std::vector<torch::Tensor>& off,
+                      int64_t H, int64_t C) {
+  const int R = (int)h_tgt.size();
+  P.n_rel = R;
+  P.cum[0] = 0;
+  for (int r = 0; r < R; ++r) {
+    TORCH_CHECK(h_tgt[r].stride(1) == C && h_tgt[r].stride(2) == 1 &&
+                    h_src[r].stride(1) == C && h_src[r].stride(2) == 1,
+                "gat_multi: inner dims must be dense");
+    P.h_tgt[r] = reinterpret_cast<const T*>(h_tgt[r].data_ptr());
+    P.h_src[r] = reinterpret_cast<const T*>(h_src[r].data_ptr());
+    P.tgt_stride[r] = h_tgt[r].stride(0);
+    P.src_stride[r] = h_src[r].stride(0);
+    P.att_src[r] = att_src[r].data_ptr<float>();
+    P.att_dst[r] = att_dst[r].data_ptr<float>();
+    P.src[r] = src[r].data_ptr<int64_t>();
+    P.off[r] = off[r].data_ptr<int64_t>();
+    P.n_tgt[r] = off[r].numel() - 1;
+    P.cum[r + 1] = P.cum[r] + P.n_tgt[r];
+  }
+}
+
+}  // namespace
+
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
+hip_gat_multi_fwd(const std::vector<torch::Tensor>& h_tgt,
+                  const std::vector<torch::Tensor>& h_src,
+                  const std::vector<torch::Tensor>& att_src,
+                  const std::vector<torch::Tensor>& att_dst,
+                  const std::vector<torch::Tensor>& src,
+                  const std::vector<torch::Tensor>& off, double slope) {
+  const int R = (int)h_tgt.size();
+  TORCH_CHECK(R >= 1 && R <= kMaxRel, "1..8 relations supported");
+  const int64_t H = h_src[0].size(1), C = h_src[0].size(2);
+  TORCH_CHECK(C <= 2 * kWave, "GAT fused kernel supports C <= 128");
+  int64_t nt_tot = 0, e_tot = 0;
+  for (int r = 0; r < R; ++r) {
+    nt_tot += off[r].numel() - 1;
+    e_tot += src[r].numel();
+  }
+  const bool bf16 = h_src[0].scalar_type() == torch::kBFloat16;
+  auto fopt = h_src[0].options().dtype(torch::kFloat32);
+  auto out = torch::empty({nt_tot, H, C}, h_src[0].options());
+  auto m = torch::empty({nt_tot, H}, fopt);
+  auto z = torch::empty({nt_tot, H}, fopt);
+  auto spre = torch::empty({e_tot, H}, fopt);
+  auto run = [&](auto type_tag) {
+    using T = decltype(type_tag);
+    GatPack<T> P;
+    fill_pack_common<T>(P, h_tgt, h_src, att_src, att_dst, src, off, H, C);
+    int64_t nt_off = 0, e_off = 0;
+    for (int r = 0; r < R; ++r) {
+      P.out[r] = reinterpret_cast<T*>(out.data_ptr()) + nt_off * H * C;
+      P.m[r] = m.data_ptr<float>() + nt_off * H;
+      P.z[r] = z.data_ptr<float>() + nt_off * H;
+      P.spre[r] = spre.data_ptr<float>() + e_off * H;
+      nt_off += P.n_tgt[r];
+      e_off += src[r].numel();
+    }
+    if (nt_tot > 0) {
+      hipLaunchKernelGGL(gat_multi_fwd_kernel<T>,
+                         dim3(gat_grid(nt_tot * H)), dim3(kBlock), 0,
+                         current_stream(), P, H, C, (float)slope, nt_tot);
+    }
+  };
+  if (bf16) run(__bf16{}); else run(float{});
+  return {out, m, z, spre};
+}
+
+void hip_gat_multi_bwd(const std::vector<torch::Tensor>& h_tgt,
+                       const std::vector<torch::Tensor>& h_src,
+                       const std::vector<torch::Tensor>& att_src,
+                       const std::vector<torch::Tensor>& att_dst,
+                       const std::vector<torch::Tensor>& src,
+                       const std::vector<torch::Tensor>& off,
+                       const torch::Tensor& out, const torch::Tensor& m,
+                       const torch::Tensor& z, const torch::Tensor& spre,
+                       const torch::Tensor& dout,
+                       const std::vector<torch::Tensor>& dh_tgt,
+                       const std::vector<torch::Tensor>& dh_src,
+                       const std::vector<torch::Tensor>& datt_src,
+                       const std::vector<torch::Tensor>& datt_dst,
+                       double slope) {
+  const int R = (int)h_tgt.size();
+  const int64_t H = h_src[0].size(1), C = h_src[0].size(2);
+  int64_t nt_tot = 0;
+  for (int r = 0; r < R; ++r) nt_tot += off[r].numel() - 1;
+  const bool bf16 = h_src[0].scalar_type() == torch::kBFloat16;
+  auto dc = dout.contiguous();
+  auto run = [&](auto type_tag) {
+    using T = decltype(type_tag);
+    GatPack<T> P;
+    fill_pack_common<T>(P, h_tgt, h_src, att_src, att_dst, src, off, H, C);
+    int64_t nt_off = 0, e_off = 0;
+    for (int r = 0; r < R; ++r) {
+      P.out[r] = reinterpret_cast<T*>(out.data_ptr()) + nt_off * H * C;
+      P.dout[r] = reinterpret_cast<const T*>(dc.data_ptr()) +
+                  nt_off * H * C;
+      P.m[r] = const_cast<float*>(m.data_ptr<float>()) + nt_off * H;
+      P.z[r] = const_cast<float*>(z.data_ptr<float>()) + nt_off * H;
+      P.spre[r] = const_cast<float*>(spre.data_ptr<float>()) + e_off * H;
+      TORCH_CHECK(dh_tgt[r].stride(1) == C && dh_src[r].stride(1) == C,
+                  "gat_multi_bwd: dh inner dims must be dense");
+      P.dh_tgt[r] = dh_tgt[r].data_ptr<float>();
+      P.dh_src[r] = dh_src[r].data_ptr<float>();
+      TORCH_CHECK(dh_tgt[r].stride(0) == P.tgt_stride[r] &&
+                      dh_src[r].stride(0) == P.src_stride[r],
+                  "gat_multi_bwd: dh strides must match h");
+      P.datt_src[r] = datt_src[r].data_ptr<float>();
+      P.datt_dst[r] = datt_dst[r].data_ptr<float>();
+      nt_off += P.n_tgt[r];
+      e_off += src[r].numel();
+    }
+    if (nt_tot > 0) {
+      int64_t S = 32768 / std::max<int64_t>(nt_tot * H, 1);
+      S = std::max<int64_t>(1, std::min<int64_t>(S, 8));
+      hipLaunchKernelGGL(gat_multi_bwd_kernel<T>,
+                         dim3(gat_grid(nt_tot * H * S)), dim3(kBlock), 0,
+                         current_stream(), P, H, C, (float)slope, S,
+                         nt_tot);
+    }
+  };
+  if (bf16) run(__bf16{}); else run(float{});
 }
 
 }  // namespace glt

@@ -112,6 +112,30 @@ hip_gat_fused_bwd(const torch::Tensor& h_tgt, const torch::Tensor& h_src,
                   const torch::Tensor& spre, const torch::Tensor& dout,
                   double slope);
 
+// Multi-relation fused GAT: one launch per hetero layer; h views may be
+// row-strided slices of the per-type batched projection.
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
+hip_gat_multi_fwd(const std::vector<torch::Tensor>& h_tgt,
+                  const std::vector<torch::Tensor>& h_src,
+                  const std::vector<torch::Tensor>& att_src,
+                  const std::vector<torch::Tensor>& att_dst,
+                  const std::vector<torch::Tensor>& src,
+                  const std::vector<torch::Tensor>& off, double slope);
+void hip_gat_multi_bwd(const std::vector<torch::Tensor>& h_tgt,
+                       const std::vector<torch::Tensor>& h_src,
+                       const std::vector<torch::Tensor>& att_src,
+                       const std::vector<torch::Tensor>& att_dst,
+                       const std::vector<torch::Tensor>& src,
+                       const std::vector<torch::Tensor>& off,
+                       const torch::Tensor& out, const torch::Tensor& m,
+                       const torch::Tensor& z, const torch::Tensor& spre,
+                       const torch::Tensor& dout,
+                       const std::vector<torch::Tensor>& dh_tgt,
+                       const std::vector<torch::Tensor>& dh_src,
+                       const std::vector<torch::Tensor>& datt_src,
+                       const std::vector<torch::Tensor>& datt_dst,
+                       double slope);
+
 // --- f32 MFMA projection GEMM (hip_gemm_f32.hip) ----------------------------
 torch::Tensor hip_sage_gemm(const torch::Tensor& A, const torch::Tensor& B,
                             const c10::optional<torch::Tensor>& bias,

@@ -53,3 +53,105 @@ def gat_softmax_aggregate(h_tgt, h_src, att_src, att_dst, tgt, src, n_tgt,
     offsets = torch.searchsorted(tgt, _boundaries(n_tgt, tgt.device))
     return _GatFused.apply(h_tgt, h_src, att_src, att_dst,
                            src.contiguous(), offsets, negative_slope)
+
+
+class _GatFusedMulti(torch.autograd.Function):
+    """ONE fused attention launch per hetero layer across relations.
+
+    Inputs are the per-TYPE batched projection tensors [N_t, R_t*H*C]
+    (consumed in place via strided views — no per-relation .contiguous()
+    copies) plus the per-relation attention parameters; the backward
+    emits ONE dh arena per type (relations accumulate atomically), so
+    the per-relation slice-grad zeros/adds disappear entirely.
+    """
+
+    @staticmethod
+    def forward(ctx, slope, spec, n_rel, n_type, *tensors):
+        from .. import _C
+
+        att_src = list(tensors[:n_rel])
+        att_dst = list(tensors[n_rel:2 * n_rel])
+        h_types = list(tensors[2 * n_rel:2 * n_rel + n_type])
+        rest = tensors[2 * n_rel + n_type:]
+        srcs = list(rest[:n_rel])
+        offs = list(rest[n_rel:2 * n_rel])
+        H, C = spec["H"], spec["C"]
+        ht_views, hs_views = [], []
+        as_f, ad_f = [], []
+        for r in range(n_rel):
+            tt, c0t, st, c0s = spec["rels"][r]
+            ht = h_types[tt]
+            hs = h_types[st]
+            ht_views.append(ht[:, c0t:c0t + H * C].view(ht.size(0), H, C))
+            hs_views.append(hs[:, c0s:c0s + H * C].view(hs.size(0), H, C))
+            as_f.append(att_src[r].reshape(H, C).float().contiguous())
+            ad_f.append(att_dst[r].reshape(H, C).float().contiguous())
+        out, m, z, spre = _C.gat_multi_fwd(ht_views, hs_views, as_f, ad_f,
+                                           srcs, offs, slope)
+        ctx.save_for_backward(out, m, z, spre, *tensors)
+        ctx.meta = (slope, spec, n_rel, n_type)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        from .. import _C
+
+        out, m, z, spre, *tensors = ctx.saved_tensors
+        slope, spec, n_rel, n_type = ctx.meta
+        att_src = list(tensors[:n_rel])
+        att_dst = list(tensors[n_rel:2 * n_rel])
+        h_types = list(tensors[2 * n_rel:2 * n_rel + n_type])
+        rest = tensors[2 * n_rel + n_type:]
+        srcs = list(rest[:n_rel])
+        offs = list(rest[n_rel:2 * n_rel])
+        H, C = spec["H"], spec["C"]
+        dh_arenas = [torch.zeros(h.size(), dtype=torch.float32,
+                                 device=h.device) for h in h_types]
+        das = torch.zeros(n_rel, H, C, dtype=torch.float32,
+                          device=out.device)
+        dad = torch.zeros_like(das)
+        ht_views, hs_views, dht_views, dhs_views = [], [], [], []
+        as_f, ad_f, das_v, dad_v = [], [], [], []
+        for r in range(n_rel):
+            tt, c0t, st, c0s = spec["rels"][r]
+            ht, hs = h_types[tt], h_types[st]
+            ht_views.append(ht[:, c0t:c0t + H * C].view(ht.size(0), H, C))
+            hs_views.append(hs[:, c0s:c0s + H * C].view(hs.size(0), H, C))
+            dht_views.append(
+                dh_arenas[tt][:, c0t:c0t + H * C].view(ht.size(0), H, C))
+            dhs_views.append(
+                dh_arenas[st][:, c0s:c0s + H * C].view(hs.size(0), H, C))
+            as_f.append(att_src[r].reshape(H, C).float().contiguous())
+            ad_f.append(att_dst[r].reshape(H, C).float().contiguous())
+            das_v.append(das[r])
+            dad_v.append(dad[r])
+        _C.gat_multi_bwd(ht_views, hs_views, as_f, ad_f, srcs, offs,
+                         out, m, z, spre, dout, dht_views, dhs_views,
+                         das_v, dad_v, slope)
+        grads = []
+        for r in range(n_rel):
+            grads.append(das[r].reshape(att_src[r].shape)
+                         .to(att_src[r].dtype))
+        for r in range(n_rel):
+            grads.append(dad[r].reshape(att_dst[r].shape)
+                         .to(att_dst[r].dtype))
+        for t in range(n_type):
+            grads.append(dh_arenas[t].to(h_types[t].dtype))
+        grads.extend([None] * (2 * n_rel))  # srcs, offs
+        return (None, None, None, None, *grads)
+
+
+def gat_multi_layer(slope, spec, att_src_list, att_dst_list, h_type_list,
+                    src_list, off_list):
+    """Run every relation's edge-softmax attention in one launch.
+
+    spec: {"H": heads, "C": channels, "rels": [(tgt_type_idx,
+    tgt_col_off, src_type_idx, src_col_off), ...]}.  Returns the
+    concatenated
+    [sum n_tgt_r, H, C] output arena (split it with torch.split).
+    """
+    n_rel = len(att_src_list)
+    n_type = len(h_type_list)
+    return _GatFusedMulti.apply(
+        slope, spec, n_rel, n_type,
+        *att_src_list, *att_dst_list, *h_type_list, *src_list, *off_list)

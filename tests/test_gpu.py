@@ -971,3 +971,53 @@ def test_deferred_sampler_empty_seeds_gpu():
     # and a seed batch of out-of-range + empty mix still guards
     out2 = s._sample_from_nodes(torch.tensor([5, -1], device="cuda"))
     assert out2.node.numel() >= 0  # no crash; guards clamp
+
+
+def test_hetero_multi_gat_matches_per_relation():
+    """The single-launch multi-relation GAT layer must match the
+    per-relation fused path bit-for-tolerance in outputs AND gradients
+    (params + inputs) on a small hetero batch."""
+    from glt_amd.models.hetero import RGNN
+
+    for dtype in (torch.float32, torch.bfloat16):
+        torch.manual_seed(4)
+        ets = [("p", "cites", "p"), ("p", "rev_writes", "a"),
+               ("a", "aff", "i")]
+        model = RGNN(ets, 32, 64, 5, num_layers=2, n_heads=2,
+                     model="rgat", dropout=0.0).cuda()
+        n = {"p": 60, "a": 30, "i": 10}
+        x = {t: torch.randn(n[t], 32, device="cuda").to(dtype)
+             for t in n}
+        ei = {}
+        for (s_t, r, d_t) in ets:
+            E = 4 * n[s_t]
+            tgt = torch.sort(torch.randint(0, n[s_t], (E,),
+                                           device="cuda")).values
+            src = torch.randint(0, n[d_t], (E,), device="cuda")
+            ei[(s_t, r, d_t)] = torch.stack([tgt, src])
+
+        def run(use_multi):
+            torch.manual_seed(9)
+            for layer in model.layers:
+                layer.use_multi = use_multi
+            for prm in model.parameters():
+                prm.grad = None
+            out = model(x, ei, predict_type="p")
+            out.float().pow(2).sum().backward()
+            return out.detach().float(), \
+                [None if prm.grad is None else prm.grad.clone()
+                 for prm in model.parameters()]
+
+        o_multi, g_multi = run(True)
+        o_single, g_single = run(False)
+        tol = 1e-4 if dtype == torch.float32 else 0.15
+        assert (o_multi - o_single).abs().max() < tol, \
+            (dtype, (o_multi - o_single).abs().max().item())
+        for gm, gs in zip(g_multi, g_single):
+            if gm is None or gs is None:
+                assert gm is None and gs is None
+                continue
+            scale = max(gs.abs().max().item(), 1.0)
+            gtol = 1e-4 if dtype == torch.float32 else 0.08
+            assert (gm - gs).abs().max() / scale < gtol, \
+                (dtype, (gm - gs).abs().max().item(), scale)
