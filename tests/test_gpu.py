@@ -1015,7 +1015,11 @@ def test_hetero_multi_gat_matches_per_relation():
             (dtype, (o_multi - o_single).abs().max().item())
         for gm, gs in zip(g_multi, g_single):
             if gm is None or gs is None:
-                assert gm is None and gs is None
+                # None vs zeros are autograd-equivalent: the multi
+                # Function materializes zero grads for structurally
+                # unused relations where autograd prunes to None
+                other = gs if gm is None else gm
+                assert other is None or not other.count_nonzero()
                 continue
             scale = max(gs.abs().max().item(), 1.0)
             gtol = 1e-4 if dtype == torch.float32 else 0.08
