@@ -283,6 +283,8 @@ struct GatPack {
   const T* dout[kMaxRel];       // bwd only (contiguous slices)
   float* datt_src[kMaxRel];
   float* datt_dst[kMaxRel];
+  const float* bias[kMaxRel];   // fwd: optional [H*C] bias per relation
+  float* dbias[kMaxRel];        // bwd: its grad (atomic colsum of dout)
   int64_t n_tgt[kMaxRel];
   int64_t cum[kMaxRel + 1];     // cumulative n_tgt (work-item decode)
   int n_rel;
@@ -358,8 +360,12 @@ __global__ void gat_multi_fwd_kernel(GatPack<T> P, int64_t H, int64_t C,
     }
     const float inv = Z > 0.f ? 1.f / Z : 0.f;
     T* ov = P.out[r] + (t * H + h) * C;
-    if (lane < C) ov[lane] = (T)(acc0 * inv);
-    if (kWave + lane < C) ov[kWave + lane] = (T)(acc1 * inv);
+    const float* bv = P.bias[r];
+    if (lane < C)
+      ov[lane] = (T)(acc0 * inv + (bv ? bv[h * C + lane] : 0.f));
+    if (kWave + lane < C)
+      ov[kWave + lane] =
+          (T)(acc1 * inv + (bv ? bv[h * C + kWave + lane] : 0.f));
     if (lane == 0) {
       P.m[r][t * H + h] = m;
       P.z[r][t * H + h] = Z;
@@ -383,6 +389,16 @@ __global__ void gat_multi_bwd_kernel(GatPack<T> P, int64_t H, int64_t C,
     int r = 0;
     while (r + 1 < P.n_rel && tg >= P.cum[r + 1]) ++r;
     const int64_t t = tg - P.cum[r];
+    if (P.dbias[r] != nullptr && q0 == 0) {
+      // d(bias) = colsum of dout over ALL targets (bias reaches empty
+      // segments too, so this runs before the empty-segment skips)
+      const T* dvb = P.dout[r] + (t * H + h) * C;
+      if (lane < C)
+        atomicAdd(&P.dbias[r][h * C + lane], (float)dvb[lane]);
+      if (kWave + lane < C)
+        atomicAdd(&P.dbias[r][h * C + kWave + lane],
+                  (float)dvb[kWave + lane]);
+    }
     const int64_t f0 = P.off[r][t], f1 = P.off[r][t + 1];
     if (f1 <= f0) continue;
     const int64_t per = (f1 - f0 + S - 1) / S;
@@ -622,7 +638,8 @@ hip_gat_multi_fwd(const std::vector<torch::Tensor>& h_tgt,
                   const std::vector<torch::Tensor>& att_src,
                   const std::vector<torch::Tensor>& att_dst,
                   const std::vector<torch::Tensor>& src,
-                  const std::vector<torch::Tensor>& off, double slope) {
+                  const std::vector<torch::Tensor>& off, double slope,
+                  const std::vector<torch::Tensor>& bias) {
   const int R = (int)h_tgt.size();
   TORCH_CHECK(R >= 1 && R <= kMaxRel, "1..8 relations supported");
   const int64_t H = h_src[0].size(1), C = h_src[0].size(2);
@@ -648,6 +665,11 @@ hip_gat_multi_fwd(const std::vector<torch::Tensor>& h_tgt,
       P.m[r] = m.data_ptr<float>() + nt_off * H;
       P.z[r] = z.data_ptr<float>() + nt_off * H;
       P.spre[r] = spre.data_ptr<float>() + e_off * H;
+      P.bias[r] = (r < (int)bias.size() && bias[r].defined() &&
+                   bias[r].numel() > 0)
+                      ? bias[r].data_ptr<float>()
+                      : nullptr;
+      P.dbias[r] = nullptr;
       nt_off += P.n_tgt[r];
       e_off += src[r].numel();
     }
@@ -674,7 +696,8 @@ void hip_gat_multi_bwd(const std::vector<torch::Tensor>& h_tgt,
                        const std::vector<torch::Tensor>& dh_src,
                        const std::vector<torch::Tensor>& datt_src,
                        const std::vector<torch::Tensor>& datt_dst,
-                       double slope) {
+                       double slope,
+                       const std::vector<torch::Tensor>& dbias) {
   const int R = (int)h_tgt.size();
   const int64_t H = h_src[0].size(1), C = h_src[0].size(2);
   int64_t nt_tot = 0;
@@ -702,6 +725,11 @@ void hip_gat_multi_bwd(const std::vector<torch::Tensor>& h_tgt,
                   "gat_multi_bwd: dh strides must match h");
       P.datt_src[r] = datt_src[r].data_ptr<float>();
       P.datt_dst[r] = datt_dst[r].data_ptr<float>();
+      P.bias[r] = nullptr;
+      P.dbias[r] = (r < (int)dbias.size() && dbias[r].defined() &&
+                    dbias[r].numel() > 0)
+                       ? dbias[r].data_ptr<float>()
+                       : nullptr;
       nt_off += P.n_tgt[r];
       e_off += src[r].numel();
     }

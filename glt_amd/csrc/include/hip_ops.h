@@ -120,7 +120,8 @@ hip_gat_multi_fwd(const std::vector<torch::Tensor>& h_tgt,
                   const std::vector<torch::Tensor>& att_src,
                   const std::vector<torch::Tensor>& att_dst,
                   const std::vector<torch::Tensor>& src,
-                  const std::vector<torch::Tensor>& off, double slope);
+                  const std::vector<torch::Tensor>& off, double slope,
+                  const std::vector<torch::Tensor>& bias);
 void hip_gat_multi_bwd(const std::vector<torch::Tensor>& h_tgt,
                        const std::vector<torch::Tensor>& h_src,
                        const std::vector<torch::Tensor>& att_src,
@@ -134,7 +135,8 @@ void hip_gat_multi_bwd(const std::vector<torch::Tensor>& h_tgt,
                        const std::vector<torch::Tensor>& dh_src,
                        const std::vector<torch::Tensor>& datt_src,
                        const std::vector<torch::Tensor>& datt_dst,
-                       double slope);
+                       double slope,
+                       const std::vector<torch::Tensor>& dbias);
 
 // --- f32 MFMA projection GEMM (hip_gemm_f32.hip) ----------------------------
 torch::Tensor hip_sage_gemm(const torch::Tensor& A, const torch::Tensor& B,

@@ -141,7 +141,8 @@ class HeteroConv(nn.Module):
                 b = self_lins[t].bias
                 self_out[t] = sl if b is None else sl + b.to(sl.dtype)
             h_list[tidx[t]] = h
-        specs, asl, adl, srcs, offs, rel_meta = [], [], [], [], [], []
+        specs, asl, adl, bl, srcs, offs = [], [], [], [], [], []
+        rel_meta = []
         for etype, key, ei in rels:
             src_t, _, dst_t = etype
             conv = convs[key]
@@ -152,11 +153,20 @@ class HeteroConv(nn.Module):
                           tidx[dst_t], col_off[(dst_t, key)]))
             asl.append(conv.att_src)
             adl.append(conv.att_dst)
+            # bias folded into the kernel epilogue; for concat=False the
+            # per-head repeat commutes with the head mean, and repeat's
+            # backward folds the [H*C] kernel grad back to [C]
+            if conv.bias is None:
+                bl.append(torch.empty(0, device=tgt.device))
+            elif conv.concat:
+                bl.append(conv.bias)
+            else:
+                bl.append(conv.bias.repeat(heads))
             srcs.append(ei[1].contiguous())
             offs.append(off_r)
             rel_meta.append((etype, key, nt))
         spec = {"H": heads, "C": C, "rels": specs}
-        arena = gat_multi_layer(c0.negative_slope, spec, asl, adl,
+        arena = gat_multi_layer(c0.negative_slope, spec, asl, adl, bl,
                                 h_list, srcs, offs)
         out: Dict[NodeType, List[torch.Tensor]] = {}
         parts = torch.split(arena, [nt for _, _, nt in rel_meta], dim=0)
@@ -164,8 +174,6 @@ class HeteroConv(nn.Module):
             conv = convs[key]
             o = part.reshape(nt, heads * C) if conv.concat \
                 else part.mean(dim=1)
-            if conv.bias is not None:
-                o = o + conv.bias.to(o.dtype)
             out.setdefault(etype[0], []).append(o)
         return out
 

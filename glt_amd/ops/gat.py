@@ -71,13 +71,14 @@ class _GatFusedMulti(torch.autograd.Function):
 
         att_src = list(tensors[:n_rel])
         att_dst = list(tensors[n_rel:2 * n_rel])
-        h_types = list(tensors[2 * n_rel:2 * n_rel + n_type])
-        rest = tensors[2 * n_rel + n_type:]
+        biases = list(tensors[2 * n_rel:3 * n_rel])  # empty = no bias
+        h_types = list(tensors[3 * n_rel:3 * n_rel + n_type])
+        rest = tensors[3 * n_rel + n_type:]
         srcs = list(rest[:n_rel])
         offs = list(rest[n_rel:2 * n_rel])
         H, C = spec["H"], spec["C"]
         ht_views, hs_views = [], []
-        as_f, ad_f = [], []
+        as_f, ad_f, b_f = [], [], []
         for r in range(n_rel):
             tt, c0t, st, c0s = spec["rels"][r]
             ht = h_types[tt]
@@ -86,8 +87,9 @@ class _GatFusedMulti(torch.autograd.Function):
             hs_views.append(hs[:, c0s:c0s + H * C].view(hs.size(0), H, C))
             as_f.append(att_src[r].reshape(H, C).float().contiguous())
             ad_f.append(att_dst[r].reshape(H, C).float().contiguous())
+            b_f.append(biases[r].reshape(-1).float().contiguous())
         out, m, z, spre = _C.gat_multi_fwd(ht_views, hs_views, as_f, ad_f,
-                                           srcs, offs, slope)
+                                           srcs, offs, slope, b_f)
         ctx.save_for_backward(out, m, z, spre, *tensors)
         ctx.meta = (slope, spec, n_rel, n_type)
         return out
@@ -100,8 +102,9 @@ class _GatFusedMulti(torch.autograd.Function):
         slope, spec, n_rel, n_type = ctx.meta
         att_src = list(tensors[:n_rel])
         att_dst = list(tensors[n_rel:2 * n_rel])
-        h_types = list(tensors[2 * n_rel:2 * n_rel + n_type])
-        rest = tensors[2 * n_rel + n_type:]
+        biases = list(tensors[2 * n_rel:3 * n_rel])
+        h_types = list(tensors[3 * n_rel:3 * n_rel + n_type])
+        rest = tensors[3 * n_rel + n_type:]
         srcs = list(rest[:n_rel])
         offs = list(rest[n_rel:2 * n_rel])
         H, C = spec["H"], spec["C"]
@@ -110,6 +113,8 @@ class _GatFusedMulti(torch.autograd.Function):
         das = torch.zeros(n_rel, H, C, dtype=torch.float32,
                           device=out.device)
         dad = torch.zeros_like(das)
+        dbias = torch.zeros(n_rel, H * C, dtype=torch.float32,
+                            device=out.device)
         ht_views, hs_views, dht_views, dhs_views = [], [], [], []
         as_f, ad_f, das_v, dad_v = [], [], [], []
         for r in range(n_rel):
@@ -125,9 +130,11 @@ class _GatFusedMulti(torch.autograd.Function):
             ad_f.append(att_dst[r].reshape(H, C).float().contiguous())
             das_v.append(das[r])
             dad_v.append(dad[r])
+        db_v = [dbias[r] if biases[r].numel() else biases[r].float()
+                for r in range(n_rel)]
         _C.gat_multi_bwd(ht_views, hs_views, as_f, ad_f, srcs, offs,
                          out, m, z, spre, dout, dht_views, dhs_views,
-                         das_v, dad_v, slope)
+                         das_v, dad_v, slope, db_v)
         grads = []
         for r in range(n_rel):
             grads.append(das[r].reshape(att_src[r].shape)
@@ -135,14 +142,17 @@ class _GatFusedMulti(torch.autograd.Function):
         for r in range(n_rel):
             grads.append(dad[r].reshape(att_dst[r].shape)
                          .to(att_dst[r].dtype))
+        for r in range(n_rel):
+            grads.append(dbias[r].to(biases[r].dtype)
+                         if biases[r].numel() else None)
         for t in range(n_type):
             grads.append(dh_arenas[t].to(h_types[t].dtype))
         grads.extend([None] * (2 * n_rel))  # srcs, offs
         return (None, None, None, None, *grads)
 
 
-def gat_multi_layer(slope, spec, att_src_list, att_dst_list, h_type_list,
-                    src_list, off_list):
+def gat_multi_layer(slope, spec, att_src_list, att_dst_list, bias_list,
+                    h_type_list, src_list, off_list):
     """Run every relation's edge-softmax attention in one launch.
 
     spec: {"H": heads, "C": channels, "rels": [(tgt_type_idx,
@@ -154,4 +164,5 @@ def gat_multi_layer(slope, spec, att_src_list, att_dst_list, h_type_list,
     n_type = len(h_type_list)
     return _GatFusedMulti.apply(
         slope, spec, n_rel, n_type,
-        *att_src_list, *att_dst_list, *h_type_list, *src_list, *off_list)
+        *att_src_list, *att_dst_list, *bias_list, *h_type_list,
+        *src_list, *off_list)
