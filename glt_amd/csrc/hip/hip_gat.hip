@@ -389,16 +389,6 @@ __global__ void gat_multi_bwd_kernel(GatPack<T> P, int64_t H, int64_t C,
     int r = 0;
     while (r + 1 < P.n_rel && tg >= P.cum[r + 1]) ++r;
     const int64_t t = tg - P.cum[r];
-    if (P.dbias[r] != nullptr && q0 == 0) {
-      // d(bias) = colsum of dout over ALL targets (bias reaches empty
-      // segments too, so this runs before the empty-segment skips)
-      const T* dvb = P.dout[r] + (t * H + h) * C;
-      if (lane < C)
-        atomicAdd(&P.dbias[r][h * C + lane], (float)dvb[lane]);
-      if (kWave + lane < C)
-        atomicAdd(&P.dbias[r][h * C + kWave + lane],
-                  (float)dvb[kWave + lane]);
-    }
     const int64_t f0 = P.off[r][t], f1 = P.off[r][t + 1];
     if (f1 <= f0) continue;
     const int64_t per = (f1 - f0 + S - 1) / S;

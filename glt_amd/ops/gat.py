@@ -130,11 +130,17 @@ class _GatFusedMulti(torch.autograd.Function):
             ad_f.append(att_dst[r].reshape(H, C).float().contiguous())
             das_v.append(das[r])
             dad_v.append(dad[r])
-        db_v = [dbias[r] if biases[r].numel() else biases[r].float()
-                for r in range(n_rel)]
+        # bias grad via one tree-reduction per biased relation (an
+        # in-kernel atomic colsum measured catastrophically contended:
+        # every (t, h) wave hits the same H*C floats)
         _C.gat_multi_bwd(ht_views, hs_views, as_f, ad_f, srcs, offs,
                          out, m, z, spre, dout, dht_views, dhs_views,
-                         das_v, dad_v, slope, db_v)
+                         das_v, dad_v, slope, [])
+        nt_sizes = [o.numel() - 1 for o in offs]
+        dsplit = torch.split(dout, nt_sizes, dim=0)
+        for r in range(n_rel):
+            if biases[r].numel():
+                dbias[r] = dsplit[r].float().sum(dim=0).reshape(-1)
         grads = []
         for r in range(n_rel):
             grads.append(das[r].reshape(att_src[r].shape)
