@@ -17,6 +17,8 @@ def main():
     ap.add_argument("--nodes", type=int, default=2_449_029)
     ap.add_argument("--edges", type=int, default=61_859_140)
     ap.add_argument("--feat-dim", type=int, default=100)
+    ap.add_argument("--dtype", type=str, default="bf16",
+                    choices=["fp32", "bf16"])
     ap.add_argument("--hidden", type=int, default=256)
     ap.add_argument("--fanout", type=str, default="10,5")
     ap.add_argument("--batch-size", type=int, default=1024)
@@ -44,7 +46,8 @@ def main():
     ds.init_graph(edge_index=torch.stack([src, dst]).cpu(),
                   graph_mode="CUDA" if has_gpu else "CPU", num_nodes=n,
                   device=0 if has_gpu else None)
-    ds.init_node_features(torch.randn(n, args.feat_dim),
+    fdt = torch.bfloat16 if args.dtype == "bf16" else torch.float32
+    ds.init_node_features(torch.randn(n, args.feat_dim).to(fdt),
                           split_ratio=1.0 if has_gpu else 0.0,
                           device=0 if has_gpu else None, with_gpu=has_gpu)
 

@@ -130,5 +130,5 @@ def unsupervised_link_pred_loss(h: torch.Tensor,
     unsupervised GraphSAGE objective)."""
     src = h[edge_label_index[0]]
     dst = h[edge_label_index[1]]
-    logits = (src * dst).sum(-1)
+    logits = (src * dst).sum(-1).float()  # bce in fp32 for bf16 runs
     return F.binary_cross_entropy_with_logits(logits, edge_label.float())
