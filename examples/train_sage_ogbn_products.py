@@ -48,6 +48,8 @@ def load_dataset(args, device):
     mode = "CUDA" if torch.cuda.is_available() else "CPU"
     ds.init_graph(edge_index=edge_index, graph_mode=mode, num_nodes=n,
                   device=device.index)
+    if args.dtype == "bf16":
+        feats = feats.to(torch.bfloat16)
     ds.init_node_features(feats, split_ratio=args.split_ratio,
                           device=device.index,
                           with_gpu=torch.cuda.is_available())
@@ -63,6 +65,12 @@ def main():
     ap.add_argument("--fanout", type=str, default="15,10,5")
     ap.add_argument("--hidden", type=int, default=256)
     ap.add_argument("--split-ratio", type=float, default=1.0)
+    ap.add_argument("--dtype", type=str, default="bf16",
+                    choices=["fp32", "bf16"],
+                    help="bf16 (default): manual mixed precision — "
+                         "features stored bf16, bf16 MFMA GEMMs/segment "
+                         "kernels, fp32 master weights (accuracy parity: "
+                         "profiles/r02_convergence.md)")
     ap.add_argument("--lr", type=float, default=0.003)
     args = ap.parse_args()
 
@@ -87,10 +95,10 @@ def main():
             out = model(data.x, data.edge_index, data.num_sampled_nodes,
                         data.num_sampled_edges)[:data.batch_size]
             y = data.y[:data.batch_size]
-            loss = F.cross_entropy(out, y)
+            loss = F.cross_entropy(out.float(), y)
             loss.backward()
             opt.step()
-            total_loss += float(loss)
+            total_loss += loss.item()
             total_correct += int((out.argmax(-1) == y).sum())
             total += y.numel()
         if torch.cuda.is_available():
