@@ -115,7 +115,12 @@ class GCNConv(nn.Module):
         tgt, src = edge_index[0], edge_index[1]
         deg = _degree(torch.cat([tgt, src]), n).to(x.dtype)
         norm = deg.rsqrt()
-        h = self.lin(x)
+        if x.dtype != self.lin.weight.dtype:
+            from ..ops import cast_linear
+
+            h = cast_linear(x, self.lin.weight, self.lin.bias)
+        else:
+            h = self.lin(x)
         msg = h.index_select(0, src) * norm[src].unsqueeze(1)
         out = h.new_zeros(nt, h.size(1))
         out.index_add_(0, tgt, msg)

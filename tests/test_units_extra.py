@@ -577,3 +577,31 @@ def test_table_dataset_reader_error_propagates():
 
     with _pytest.raises(RuntimeError, match="boom"):
         list(stream_table("t", 2, 4, factory))
+
+
+def test_gcn_bf16_cpu():
+    from glt_amd.models.layers import GCNConv
+
+    torch.manual_seed(0)
+    conv = GCNConv(8, 8)
+    x = torch.randn(20, 8).to(torch.bfloat16).requires_grad_(True)
+    tgt = torch.randint(0, 10, (40,))
+    src = torch.randint(0, 20, (40,))
+    out = conv(x, torch.stack([tgt, src]), num_target=10)
+    assert out.dtype == torch.bfloat16
+    out.float().sum().backward()
+    assert conv.lin.weight.grad.dtype == torch.float32
+
+
+def test_bf16_mfma_escape_hatch(monkeypatch):
+    """GLT_DISABLE_BF16_MFMA forces the hipBLASLt path (ops/linear.py
+    dispatch policy)."""
+    import glt_amd.ops.linear as L
+
+    w = torch.randn(64, 32)
+    x = torch.randn(4, 32).to(torch.bfloat16)
+    monkeypatch.setenv("GLT_DISABLE_BF16_MFMA", "1")
+    assert not L._use_bf16_mfma(x, w)
+    monkeypatch.delenv("GLT_DISABLE_BF16_MFMA")
+    # (CPU tensors also refuse the device kernels)
+    assert not L._use_bf16_mfma(x, w)
