@@ -171,6 +171,126 @@ void gemm_bt_kernel(const __bf16* __restrict__ A,
   }
 }
 
+
+// Persistent-B forward variant for K <= 256: the whole Bt n-tile
+// ([64][K] bf16 <= 32 KB) is staged ONCE per workgroup, then A m-tiles
+// stream through with a single sync pair each and one full-K MFMA pass
+// — the generic kernel re-stages B and syncs per 64-K step.
+constexpr int PKP = 256;  // padded K capacity
+
+template <bool RELU, bool BF16_OUT>
+__global__ __launch_bounds__(256)
+void gemm_bt_persist_kernel(const __bf16* __restrict__ A,
+                            const __bf16* __restrict__ Bt,
+                            const float* __restrict__ bias,
+                            void* __restrict__ Cv,
+                            int64_t M, int64_t K, int64_t N) {
+  __shared__ __bf16 As[GBM][PKP + LDP];
+  __shared__ __bf16 Bs[GBN][PKP + LDP];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1;
+  const int wc = wave & 1;
+  const int64_t block_col = (int64_t)blockIdx.y * GBN;
+  const int s_row = tid >> 2;          // 0..63
+  const int s_col0 = (tid & 3) * 64;   // 4 threads cover 256 cols
+  const int fi = lane & 15;
+  const int fk8 = (lane >> 4) * 8;
+  const int kp = (int)((K + 31) / 32) * 32;
+
+  // stage Bt rows once (zero-padded K tail)
+  {
+    const int64_t g_row = block_col + s_row;
+#pragma unroll
+    for (int c8 = 0; c8 < 64; c8 += 8) {
+      const int col = s_col0 + c8;
+      bf16x8 v = {};
+      if (col < kp && g_row < N) {
+        const int64_t base = g_row * K + col;
+        if (col + 7 < K) {
+          v = *reinterpret_cast<const bf16x8*>(&Bt[base]);
+        } else {
+#pragma unroll
+          for (int e = 0; e < 8; ++e)
+            if (col + e < K) v[e] = Bt[base + e];
+        }
+      }
+      if (col < kp)
+        *reinterpret_cast<bf16x8*>(&Bs[s_row][col]) = v;
+    }
+  }
+  __syncthreads();
+
+  for (int64_t m0 = (int64_t)blockIdx.x * GBM; m0 < M;
+       m0 += (int64_t)gridDim.x * GBM) {
+    {
+      const int64_t g_row = m0 + s_row;
+#pragma unroll
+      for (int c8 = 0; c8 < 64; c8 += 8) {
+        const int col = s_col0 + c8;
+        bf16x8 v = {};
+        if (col < kp && g_row < M) {
+          const int64_t base = g_row * K + col;
+          if (col + 7 < K) {
+            v = *reinterpret_cast<const bf16x8*>(&A[base]);
+          } else {
+#pragma unroll
+            for (int e = 0; e < 8; ++e)
+              if (col + e < K) v[e] = A[base + e];
+          }
+        }
+        if (col < kp)
+          *reinterpret_cast<bf16x8*>(&As[s_row][col]) = v;
+      }
+    }
+    __syncthreads();
+    f32x4b acc[2][2] = {};
+    for (int kk = 0; kk < kp; kk += 32) {
+      bf16x8 a0 = *reinterpret_cast<const bf16x8*>(
+          &As[wr * 32 + fi][kk + fk8]);
+      bf16x8 a1 = *reinterpret_cast<const bf16x8*>(
+          &As[wr * 32 + 16 + fi][kk + fk8]);
+      bf16x8 b0 = *reinterpret_cast<const bf16x8*>(
+          &Bs[wc * 32 + fi][kk + fk8]);
+      bf16x8 b1 = *reinterpret_cast<const bf16x8*>(
+          &Bs[wc * 32 + 16 + fi][kk + fk8]);
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0,
+                                                          acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1,
+                                                          acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0,
+                                                          acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1,
+                                                          acc[1][1], 0, 0, 0);
+    }
+    __syncthreads();
+    const int c_col = lane & 15;
+    const int c_row0 = (lane >> 4) * 4;
+#pragma unroll
+    for (int fi2 = 0; fi2 < 2; ++fi2) {
+#pragma unroll
+      for (int fj = 0; fj < 2; ++fj) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int64_t row = m0 + wr * 32 + fi2 * 16 + c_row0 + r;
+          const int64_t col = block_col + wc * 32 + fj * 16 + c_col;
+          if (row < M && col < N) {
+            float v = acc[fi2][fj][r];
+            if (bias != nullptr) v += bias[col];
+            if (RELU) v = v > 0.f ? v : 0.f;
+            if (BF16_OUT)
+              reinterpret_cast<__bf16*>(Cv)[row * N + col] = f2bf(v);
+            else
+              reinterpret_cast<float*>(Cv)[row * N + col] = v;
+          }
+        }
+      }
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 // dW[M,N] = A[Kb,M]^T @ B[Kb,N] (+ db[M] = colsum A), split-K over
 // grid.z with per-chunk partial planes.  A/B tiles are transposed
@@ -515,6 +635,21 @@ torch::Tensor hip_gemm_bt_bf16(const torch::Tensor& A,
   if (bias.has_value()) {
     bias_f = bias->to(torch::kFloat32).contiguous();
     bias_p = bias_f.data_ptr<float>();
+  }
+  if (K <= PKP) {
+    const int64_t m_t = (M + GBM - 1) / GBM;
+    dim3 grid((uint32_t)std::min<int64_t>(m_t, 1024),
+              (uint32_t)((N + GBN - 1) / GBN));
+    auto* kfn = out_fp32
+                    ? (relu ? gemm_bt_persist_kernel<true, false>
+                            : gemm_bt_persist_kernel<false, false>)
+                    : (relu ? gemm_bt_persist_kernel<true, true>
+                            : gemm_bt_persist_kernel<false, true>);
+    hipLaunchKernelGGL(kfn, grid, dim3(256), 0, current_stream(),
+                       reinterpret_cast<const __bf16*>(Ac.data_ptr()),
+                       reinterpret_cast<const __bf16*>(Bc.data_ptr()),
+                       bias_p, C.data_ptr(), M, K, N);
+    return C;
   }
   dim3 grid((uint32_t)((M + GBM - 1) / GBM),
             (uint32_t)((N + GBN - 1) / GBN));
