@@ -14,9 +14,7 @@
  *     - the weight-gradient shape (K huge = batch rows): split-K over
  *       grid.z into per-chunk partial planes (a torch sum folds them);
  *       the bias gradient rides the same kernel (the separate
- *       [rows,256] bf16 column-reduce measured 78 us/call).  Runs on
- *       the f32 16x16x4 MFMA whose k-per-lane-group operand layout
- *       matches the [k][cols] inputs with no LDS transpose.
+ *       [rows,256] bf16 column-reduce measured 78 us/call).
  *     - fp32 out: master-grad dtype, so the .to(fp32) casts disappear.
  *  3. mfma_bf16_selftest: one 16x16x32 / 32x32x16 tile from explicit
  *     matrices (fragment-layout ground truth for the GPU tests).
@@ -28,6 +26,8 @@
  * K dim is innermost in global memory already), the kt kernel stages
  * transposed during the write pass instead.
  */
+#include <cstdlib>
+
 #include "hip_common.h"
 #include "../include/common.h"
 
@@ -636,7 +636,13 @@ torch::Tensor hip_gemm_bt_bf16(const torch::Tensor& A,
     bias_f = bias->to(torch::kFloat32).contiguous();
     bias_p = bias_f.data_ptr<float>();
   }
-  if (K <= PKP) {
+  // NOTE: a persistent-B variant (whole <=256-K weight tile staged once,
+  // A streaming with one sync pair per m-tile) measured SLOWER: its
+  // 68 KB LDS footprint halves occupancy (2 WGs/CU vs 4+) and the lost
+  // latency hiding on the A stream outweighs the saved syncs/B-restage
+  // (L0 fwd 167 us vs 112; flagship 661 vs 720 b/s).  Enable with
+  // GLT_GEMM_PERSIST=1 for further experiments.
+  if (K <= PKP && std::getenv("GLT_GEMM_PERSIST") != nullptr) {
     const int64_t m_t = (M + GBM - 1) / GBM;
     dim3 grid((uint32_t)std::min<int64_t>(m_t, 1024),
               (uint32_t)((N + GBN - 1) / GBN));
