@@ -18,14 +18,23 @@ from .transform import to_data, to_hetero_data
 class _SeedIterator:
     def __init__(self, seeds: torch.Tensor, batch_size: int, shuffle: bool,
                  drop_last: bool, generator=None):
-        self.seeds = seeds
         self.batch_size = batch_size
         self.drop_last = drop_last
         n = seeds.numel()
         if shuffle:
-            self.order = torch.randperm(n, generator=generator)
-        else:
-            self.order = None
+            # materialize the permutation once: batches become
+            # contiguous SLICES, which stay pinned when the seed tensor
+            # is pinned (per-batch fancy indexing would return unpinned
+            # copies and turn every seed upload into a blocking H2D)
+            seeds = seeds[torch.randperm(n, generator=generator)
+                          .to(seeds.device)]
+            if not seeds.is_cuda and torch.cuda.is_available():
+                try:
+                    seeds = seeds.pin_memory()
+                except RuntimeError:
+                    pass
+        self.seeds = seeds
+        self.order = None
         self.pos = 0
         self.n = n
 
@@ -140,6 +149,16 @@ class NodeLoader:
         self.to_device = to_device
         self.prefetch = prefetch
         self._prefetcher = None
+        # pin CPU seed tensors once so per-batch seed uploads are true
+        # async H2D (pageable copies stall the producer thread; the
+        # hetero producer issues one per (hop, etype))
+        node = self.input_nodes.node
+        if (torch.cuda.is_available() and torch.is_tensor(node)
+                and not node.is_cuda and not node.is_pinned()):
+            try:
+                self.input_nodes.node = node.pin_memory()
+            except RuntimeError:
+                pass
 
     def __iter__(self):
         self._it = _SeedIterator(self.input_nodes.node, self.batch_size,
