@@ -280,6 +280,8 @@ struct GatPack {
   float* spre[kMaxRel];
   float* dh_tgt[kMaxRel];       // bwd only (fp32 arenas, strided)
   float* dh_src[kMaxRel];
+  int64_t dht_stride[kMaxRel];
+  int64_t dhs_stride[kMaxRel];
   const T* dout[kMaxRel];       // bwd only (contiguous slices)
   float* datt_src[kMaxRel];
   float* datt_dst[kMaxRel];
@@ -417,6 +419,7 @@ __global__ void gat_multi_bwd_kernel(GatPack<T> P, int64_t H, int64_t C,
     float dad_acc = 0.f;
     float das0 = 0.f, das1 = 0.f;
     const int64_t hstride = P.src_stride[r];
+    const int64_t dstride = P.dhs_stride[r];
     const T* hbase = P.h_src[r] + h * C;
     float* dhbase = P.dh_src[r] + h * C;
     for (int64_t e = s0; e < s1; e += 4) {
@@ -452,7 +455,7 @@ __global__ void gat_multi_bwd_kernel(GatPack<T> P, int64_t H, int64_t C,
         const float pp = __expf(sa - m) * inv;
         float ds = pp * (__shfl(dot_h4[q], 0) - dot_o);
         ds *= (s_pre > 0.f ? 1.f : slope);
-        float* dhv = dhbase + sn[q] * hstride;
+        float* dhv = dhbase + sn[q] * dstride;
         if (lane < C) atomicAdd(&dhv[lane], pp * d0 + ds * as0);
         if (kWave + lane < C)
           atomicAdd(&dhv[kWave + lane], pp * d1 + ds * as1);
@@ -461,7 +464,7 @@ __global__ void gat_multi_bwd_kernel(GatPack<T> P, int64_t H, int64_t C,
         dad_acc += ds;
       }
     }
-    float* dtv = P.dh_tgt[r] + t * P.tgt_stride[r] + h * C;
+    float* dtv = P.dh_tgt[r] + t * P.dht_stride[r] + h * C;
     if (lane < C) {
       atomicAdd(&dtv[lane], dad_acc * ad0);
       atomicAdd(&P.datt_src[r][h * C + lane], das0);
@@ -710,9 +713,8 @@ void hip_gat_multi_bwd(const std::vector<torch::Tensor>& h_tgt,
                   "gat_multi_bwd: dh inner dims must be dense");
       P.dh_tgt[r] = dh_tgt[r].data_ptr<float>();
       P.dh_src[r] = dh_src[r].data_ptr<float>();
-      TORCH_CHECK(dh_tgt[r].stride(0) == P.tgt_stride[r] &&
-                      dh_src[r].stride(0) == P.src_stride[r],
-                  "gat_multi_bwd: dh strides must match h");
+      P.dht_stride[r] = dh_tgt[r].stride(0);
+      P.dhs_stride[r] = dh_src[r].stride(0);
       P.datt_src[r] = datt_src[r].data_ptr<float>();
       P.datt_dst[r] = datt_dst[r].data_ptr<float>();
       P.bias[r] = nullptr;

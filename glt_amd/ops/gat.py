@@ -83,14 +83,21 @@ class _GatFusedMulti(torch.autograd.Function):
             tt, c0t, st, c0s = spec["rels"][r]
             ht = h_types[tt]
             hs = h_types[st]
-            ht_views.append(ht[:, c0t:c0t + H * C].view(ht.size(0), H, C))
-            hs_views.append(hs[:, c0s:c0s + H * C].view(hs.size(0), H, C))
+            # contiguous per-relation h copies for the kernel READS
+            # (strided slices of the wide projection measured ~2.5x
+            # slower attention kernels); gradients still land in ONE
+            # strided arena per type
+            ht_views.append(ht[:, c0t:c0t + H * C]
+                            .contiguous().view(ht.size(0), H, C))
+            hs_views.append(hs[:, c0s:c0s + H * C]
+                            .contiguous().view(hs.size(0), H, C))
             as_f.append(att_src[r].reshape(H, C).float().contiguous())
             ad_f.append(att_dst[r].reshape(H, C).float().contiguous())
             b_f.append(biases[r].reshape(-1).float().contiguous())
         out, m, z, spre = _C.gat_multi_fwd(ht_views, hs_views, as_f, ad_f,
                                            srcs, offs, slope, b_f)
         ctx.save_for_backward(out, m, z, spre, *tensors)
+        ctx.h_views = (ht_views, hs_views)  # contiguous, reused by bwd
         ctx.meta = (slope, spec, n_rel, n_type)
         return out
 
@@ -115,13 +122,12 @@ class _GatFusedMulti(torch.autograd.Function):
         dad = torch.zeros_like(das)
         dbias = torch.zeros(n_rel, H * C, dtype=torch.float32,
                             device=out.device)
-        ht_views, hs_views, dht_views, dhs_views = [], [], [], []
+        ht_views, hs_views = ctx.h_views
+        dht_views, dhs_views = [], []
         as_f, ad_f, das_v, dad_v = [], [], [], []
         for r in range(n_rel):
             tt, c0t, st, c0s = spec["rels"][r]
             ht, hs = h_types[tt], h_types[st]
-            ht_views.append(ht[:, c0t:c0t + H * C].view(ht.size(0), H, C))
-            hs_views.append(hs[:, c0s:c0s + H * C].view(hs.size(0), H, C))
             dht_views.append(
                 dh_arenas[tt][:, c0t:c0t + H * C].view(ht.size(0), H, C))
             dhs_views.append(
