@@ -304,6 +304,10 @@ __global__ void gat_multi_fwd_kernel(GatPack<T> P, int64_t H, int64_t C,
     const int64_t h = w - tg * H;
     int r = 0;
     while (r + 1 < P.n_rel && tg >= P.cum[r + 1]) ++r;
+    // r is wave-uniform by construction (w is the wave index); telling
+    // the compiler so keeps every P.*[r] select scalar instead of
+    // per-lane kernarg fetches
+    r = __builtin_amdgcn_readfirstlane(r);
     const int64_t t = tg - P.cum[r];
     const int64_t s0 = P.off[r][t], s1 = P.off[r][t + 1];
     const float* ats = P.att_src[r] + h * C;
@@ -390,6 +394,7 @@ __global__ void gat_multi_bwd_kernel(GatPack<T> P, int64_t H, int64_t C,
     const int64_t q0 = rem - h * S;
     int r = 0;
     while (r + 1 < P.n_rel && tg >= P.cum[r + 1]) ++r;
+    r = __builtin_amdgcn_readfirstlane(r);
     const int64_t t = tg - P.cum[r];
     const int64_t f0 = P.off[r][t], f1 = P.off[r][t + 1];
     if (f1 <= f0) continue;
