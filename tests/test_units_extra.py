@@ -605,3 +605,18 @@ def test_bf16_mfma_escape_hatch(monkeypatch):
     monkeypatch.delenv("GLT_DISABLE_BF16_MFMA")
     # (CPU tensors also refuse the device kernels)
     assert not L._use_bf16_mfma(x, w)
+
+
+def test_feature_id2index_device_cache():
+    """id2index is cached per device — a per-lookup .to() re-uploaded
+    the full map every batch (round-1 advisor finding)."""
+    from glt_amd.data import Feature
+
+    feats = torch.arange(10, dtype=torch.float32).unsqueeze(1)
+    id2i = torch.arange(10)
+    f = Feature(feats, with_gpu=False, id2index=id2i)
+    a = f._id2index_on(torch.device("cpu"))
+    b = f._id2index_on(torch.device("cpu"))
+    assert a.data_ptr() == b.data_ptr()  # same cached tensor
+    out = f[torch.tensor([3, 7])]
+    assert out.flatten().tolist() == [3.0, 7.0]
