@@ -121,7 +121,8 @@ class HeteroConv(nn.Module):
         for t, keys in needs.items():
             x = x_dict[t]
             ws = [convs[k].lin.weight for k in keys]
-            if len({w.size(1) for w in ws}) != 1 or                     ws[0].size(1) != x.size(1):
+            if len({w.size(1) for w in ws}) != 1 or \
+                    ws[0].size(1) != x.size(1):
                 return None
             fold_self = (self_lins is not None and t in self_lins
                          and self_lins[t].weight.size(1) == x.size(1))
