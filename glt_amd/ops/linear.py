@@ -76,14 +76,20 @@ class _CastLinear(torch.autograd.Function):
             out = torch.nn.functional.linear(x, w16, b16)
             if relu:
                 out = torch.relu_(out)
-        ctx.save_for_backward(x, w16, out)
+        if relu:  # out is only needed for the ReLU mask in backward
+            ctx.save_for_backward(x, w16, out)
+        else:
+            ctx.save_for_backward(x, w16)
         ctx.has_bias = bias is not None
         ctx.relu = relu
         return out
 
     @staticmethod
     def backward(ctx, dy):
-        x, w16, out = ctx.saved_tensors
+        if ctx.relu:
+            x, w16, out = ctx.saved_tensors
+        else:
+            x, w16 = ctx.saved_tensors
         dy = dy.contiguous()
         if ctx.relu:
             # one-kernel relu mask (vs gt + mul two-kernel form)
