@@ -362,6 +362,7 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = t.item()
 
+    loader.shutdown()
     batches_per_sec = args.steps / elapsed * world
     ms_per_step = elapsed / args.steps * 1000.0
     # ogbn-products train split: 196,615 seeds -> epoch equivalent

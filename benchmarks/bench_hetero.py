@@ -125,6 +125,7 @@ def main():
     if has_gpu:
         torch.cuda.synchronize()
     dt = time.perf_counter() - t0
+    loader.shutdown()
     print(json.dumps({
         "metric": f"{args.model} IGBH-shaped hetero train batches/sec",
         "value": round(args.steps / dt, 3),

@@ -93,6 +93,7 @@ def main():
     if has_gpu:
         torch.cuda.synchronize()
     dt = time.perf_counter() - t0
+    loader.shutdown()
     print(json.dumps({
         "metric": "unsup GraphSAGE link-pred batches/sec",
         "value": round(args.steps / dt, 3),

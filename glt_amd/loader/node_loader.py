@@ -115,12 +115,18 @@ class _Prefetcher:
         return data
 
     def stop(self):
+        """Stop and JOIN the producer thread.  A daemon thread killed
+        mid-HIP-call at interpreter exit aborts the process ("terminate
+        called without an active exception" in the HIP runtime
+        teardown) — observed when the consumer drains batches faster
+        than the producer makes them (sampler-only benchmarks)."""
         self._stop = True
         try:
             while True:
                 self._queue.get_nowait()
         except Exception:
             pass
+        self._thread.join(timeout=5)
 
 
 class NodeLoader:
@@ -184,6 +190,19 @@ class NodeLoader:
         if self._prefetcher is not None:
             return self._prefetcher.next()
         return self._produce(next(self._it))
+
+    def shutdown(self):
+        """Stop the background prefetcher (called automatically on
+        garbage collection; call explicitly before interpreter exit)."""
+        if self._prefetcher is not None:
+            self._prefetcher.stop()
+            self._prefetcher = None
+
+    def __del__(self):
+        try:
+            self.shutdown()
+        except Exception:
+            pass
 
     # -- feature/label collection ------------------------------------------
     def _collate_fn(self, out: Union[SamplerOutput, HeteroSamplerOutput]):
