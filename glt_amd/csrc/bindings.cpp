@@ -159,6 +159,25 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       py::arg("edge_ids") = py::none(), py::arg("edge_weights") = py::none(),
       py::arg("with_edge") = false, py::arg("weighted") = false,
       py::arg("replace") = true);
+  m.def("sample_neighbors_offsets", &hip_sample_neighbors_offsets,
+        py::arg("indptr"), py::arg("seeds"), py::arg("k"));
+  m.def(
+      "sample_neighbors_gather",
+      [](const torch::Tensor& indptr, const torch::Tensor& indices,
+         const torch::Tensor& seeds, int64_t k,
+         const torch::Tensor& offsets, int64_t total,
+         const OptTensor& edge_ids, const OptTensor& edge_weights,
+         bool with_edge, bool weighted, bool replace) {
+        return hip_sample_neighbors_gather(indptr, indices, edge_ids,
+                                           edge_weights, seeds, k, offsets,
+                                           total, with_edge, weighted,
+                                           replace);
+      },
+      py::arg("indptr"), py::arg("indices"), py::arg("seeds"), py::arg("k"),
+      py::arg("offsets"), py::arg("total"),
+      py::arg("edge_ids") = py::none(), py::arg("edge_weights") = py::none(),
+      py::arg("with_edge") = false, py::arg("weighted") = false,
+      py::arg("replace") = true);
   m.def("lookup_degree", &lookup_degree);
   m.def("sample_negative", &sample_negative, py::arg("indptr"),
         py::arg("indices"), py::arg("num_cols"), py::arg("req_num"),

@@ -385,33 +385,51 @@ uint64_t fresh_seed() { return SeedManager::instance().next_call_seed(); }
 
 // Host entry points ---------------------------------------------------------
 
-std::tuple<torch::Tensor, torch::Tensor, c10::optional<torch::Tensor>>
-hip_sample_neighbors(const torch::Tensor& indptr, const torch::Tensor& indices,
-                     const c10::optional<torch::Tensor>& edge_ids,
-                     const c10::optional<torch::Tensor>& edge_weights,
-                     const torch::Tensor& seeds, int64_t k, bool with_edge,
-                     bool weighted, bool replace) {
-  TORCH_CHECK(indptr.is_cuda() && indices.is_cuda() && seeds.is_cuda(),
-              "hip_sample_neighbors: tensors must be on GPU (or mapped views)");
-  TORCH_CHECK(!with_edge || edge_ids.has_value(), "with_edge requires edge_ids");
+// --- staged sampling API ----------------------------------------------------
+// Stage 1: per-seed counts + exclusive offsets, NO host sync.  The
+// hetero multihop runs stage 1 for every edge type of a hop, reads all
+// the totals with ONE .cpu() round trip, then runs stage 2 — one
+// GIL-held device sync per hop instead of one per (hop, etype), which
+// is what throttled the in-process producer thread (ROUND2_NOTES).
+std::tuple<torch::Tensor, torch::Tensor> hip_sample_neighbors_offsets(
+    const torch::Tensor& indptr, const torch::Tensor& seeds, int64_t k) {
+  TORCH_CHECK(indptr.is_cuda() && seeds.is_cuda(),
+              "sample_neighbors_offsets: device tensors required");
+  const int64_t num_rows = indptr.size(0) - 1;
+  const int64_t bs = seeds.size(0);
+  const int64_t kk = k < 0 ? std::numeric_limits<int64_t>::max() : k;
+  auto counts = torch::empty({bs}, seeds.options());
+  if (bs > 0) {
+    hipLaunchKernelGGL(fill_counts_kernel, dim3(grid_for(bs)), dim3(kBlock),
+                       0, current_stream(), indptr.data_ptr<int64_t>(),
+                       num_rows, seeds.data_ptr<int64_t>(), bs, kk,
+                       counts.data_ptr<int64_t>());
+  }
+  auto offsets = torch::zeros({bs + 1}, seeds.options());
+  if (bs > 0) {
+    auto v = offsets.narrow(0, 1, bs);
+    torch::cumsum_out(v, counts, 0);
+  }
+  return {counts, offsets};
+}
+
+// Stage 2: gather with the caller-synced total.
+std::tuple<torch::Tensor, c10::optional<torch::Tensor>>
+hip_sample_neighbors_gather(const torch::Tensor& indptr,
+                            const torch::Tensor& indices,
+                            const c10::optional<torch::Tensor>& edge_ids,
+                            const c10::optional<torch::Tensor>& edge_weights,
+                            const torch::Tensor& seeds, int64_t k,
+                            const torch::Tensor& offsets, int64_t total,
+                            bool with_edge, bool weighted, bool replace) {
+  TORCH_CHECK(!with_edge || edge_ids.has_value(),
+              "with_edge requires edge_ids");
   TORCH_CHECK(!weighted || edge_weights.has_value(),
               "weighted requires edge_weights");
   const int64_t num_rows = indptr.size(0) - 1;
   const int64_t bs = seeds.size(0);
   const int64_t kk = k < 0 ? std::numeric_limits<int64_t>::max() : k;
   auto stream = current_stream();
-
-  auto counts = torch::empty({bs}, seeds.options());
-  if (bs > 0) {
-    hipLaunchKernelGGL(fill_counts_kernel, dim3(grid_for(bs)), dim3(kBlock), 0,
-                       stream, indptr.data_ptr<int64_t>(), num_rows,
-                       seeds.data_ptr<int64_t>(), bs, kk,
-                       counts.data_ptr<int64_t>());
-  }
-  auto offsets = torch::zeros({bs + 1}, seeds.options());
-  { auto v = offsets.narrow(0, 1, bs); torch::cumsum_out(v, counts, 0); }
-  const int64_t total = bs > 0 ? offsets[bs].item<int64_t>() : 0;  // hop sync
-
   auto nbrs = torch::empty({total}, seeds.options());
   auto out_eids = with_edge ? torch::empty({total}, seeds.options())
                             : torch::Tensor();
@@ -483,8 +501,26 @@ hip_sample_neighbors(const torch::Tensor& indptr, const torch::Tensor& indices,
       }
     }
   }
-  return {nbrs, counts,
+  return {nbrs,
           with_edge ? c10::optional<torch::Tensor>(out_eids) : c10::nullopt};
+}
+
+std::tuple<torch::Tensor, torch::Tensor, c10::optional<torch::Tensor>>
+hip_sample_neighbors(const torch::Tensor& indptr, const torch::Tensor& indices,
+                     const c10::optional<torch::Tensor>& edge_ids,
+                     const c10::optional<torch::Tensor>& edge_weights,
+                     const torch::Tensor& seeds, int64_t k, bool with_edge,
+                     bool weighted, bool replace) {
+  TORCH_CHECK(indptr.is_cuda() && indices.is_cuda() && seeds.is_cuda(),
+              "hip_sample_neighbors: tensors must be on GPU (or mapped views)");
+  auto [counts, offsets] = hip_sample_neighbors_offsets(indptr, seeds, k);
+  const int64_t bs = seeds.size(0);
+  const int64_t total =
+      bs > 0 ? offsets[bs].item<int64_t>() : 0;  // hop sync
+  auto [nbrs, out_eids] = hip_sample_neighbors_gather(
+      indptr, indices, edge_ids, edge_weights, seeds, k, offsets, total,
+      with_edge, weighted, replace);
+  return {nbrs, counts, out_eids};
 }
 
 torch::Tensor hip_lookup_degree(const torch::Tensor& indptr,

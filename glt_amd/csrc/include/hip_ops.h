@@ -21,6 +21,18 @@ hip_sample_neighbors(const torch::Tensor& indptr, const torch::Tensor& indices,
                      const c10::optional<torch::Tensor>& edge_weights,
                      const torch::Tensor& seeds, int64_t k, bool with_edge,
                      bool weighted, bool replace = true);
+// Staged sampling: stage 1 (counts+offsets, no host sync) lets callers
+// batch the totals sync across edge types; stage 2 gathers.
+std::tuple<torch::Tensor, torch::Tensor> hip_sample_neighbors_offsets(
+    const torch::Tensor& indptr, const torch::Tensor& seeds, int64_t k);
+std::tuple<torch::Tensor, c10::optional<torch::Tensor>>
+hip_sample_neighbors_gather(const torch::Tensor& indptr,
+                            const torch::Tensor& indices,
+                            const c10::optional<torch::Tensor>& edge_ids,
+                            const c10::optional<torch::Tensor>& edge_weights,
+                            const torch::Tensor& seeds, int64_t k,
+                            const torch::Tensor& offsets, int64_t total,
+                            bool with_edge, bool weighted, bool replace);
 torch::Tensor hip_lookup_degree(const torch::Tensor& indptr,
                                 const torch::Tensor& nodes);
 torch::Tensor hip_sample_negative(const torch::Tensor& indptr,
