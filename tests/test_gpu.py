@@ -1025,3 +1025,26 @@ def test_hetero_multi_gat_matches_per_relation():
             gtol = 1e-4 if dtype == torch.float32 else 0.08
             assert (gm - gs).abs().max() / scale < gtol, \
                 (dtype, (gm - gs).abs().max().item(), scale)
+
+
+def test_staged_sampling_matches_classic():
+    """sample_neighbors_offsets + _gather must reproduce the one-call
+    sampler BIT-IDENTICALLY under the same seed (the hetero multihop
+    relies on this to batch its per-hop totals sync)."""
+    glt_amd.seed_everything(5)
+    n = 5000
+    src = torch.randint(0, n, (60_000,))
+    dst = torch.randint(0, n, (60_000,))
+    topo = Topology(torch.stack([src, dst]), num_nodes=n)
+    indptr, indices = topo.indptr.cuda(), topo.indices.cuda()
+    seeds = torch.randint(0, n, (777,), device="cuda")
+
+    glt_amd.seed_everything(7)
+    nbrs1, num1, _ = _C.sample_neighbors(indptr, indices, seeds, 7)
+    glt_amd.seed_everything(7)
+    counts, offs = _C.sample_neighbors_offsets(indptr, seeds, 7)
+    total = int(offs[-1].item())
+    nbrs2, _ = _C.sample_neighbors_gather(indptr, indices, seeds, 7,
+                                          offs, total)
+    assert torch.equal(num1, counts)
+    assert torch.equal(nbrs1, nbrs2)
