@@ -101,6 +101,18 @@ struct DeviceInducer {
   torch::Tensor insert(const torch::Tensor& ids) {
     return hip_inducer_insert(p.get(), ids);
   }
+  void reserve_incoming(int64_t total) {
+    hip_inducer_reserve(p.get(), total);
+  }
+  std::tuple<torch::Tensor, torch::Tensor> insert_begin(
+      const torch::Tensor& ids, int64_t idx_base) {
+    return hip_inducer_insert_begin(p.get(), ids, idx_base);
+  }
+  torch::Tensor insert_commit(const torch::Tensor& ids,
+                              const torch::Tensor& flags,
+                              const torch::Tensor& ranks, int64_t n_new) {
+    return hip_inducer_insert_commit(p.get(), ids, flags, ranks, n_new);
+  }
   int64_t count() { return hip_inducer_count(p.get()); }
 };
 
@@ -215,6 +227,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("induce_next", &DeviceInducer::induce_next)
       .def("lookup", &DeviceInducer::lookup)
       .def("insert", &DeviceInducer::insert)
+      .def("reserve_incoming", &DeviceInducer::reserve_incoming)
+      .def("insert_begin", &DeviceInducer::insert_begin)
+      .def("insert_commit", &DeviceInducer::insert_commit)
       .def("count", &DeviceInducer::count);
 
   py::class_<DeferredSamplerPy>(m, "DeferredSampler")

@@ -68,23 +68,30 @@ __device__ __forceinline__ int64_t table_lookup(TableView t, uint64_t key) {
   }
 }
 
+// idx_base: staged multi-call insertion (one hop's edge types insert
+// back-to-back BEFORE any assign) biases the first-occurrence order key
+// so later calls never steal first place from earlier ones — keeps the
+// fresh-node order identical to the classic one-sync-per-call flow.
 __global__ void insert_kernel(TableView t, const int64_t* __restrict__ ids,
-                              int64_t n) {
+                              int64_t n, int64_t idx_base) {
   for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
-    table_insert(t, (uint64_t)ids[i], (uint64_t)i);
+    table_insert(t, (uint64_t)ids[i], (uint64_t)(idx_base + i));
   }
 }
 
 // flag[i] = 1 iff position i is the first occurrence of ids[i].
 __global__ void flag_first_kernel(TableView t, const int64_t* __restrict__ ids,
-                                  int64_t n, int64_t* __restrict__ flags) {
+                                  int64_t n, int64_t idx_base,
+                                  int64_t* __restrict__ flags) {
   for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     const uint64_t slot = table_find(t, (uint64_t)ids[i]);
     // Only fresh nodes count: a slot owned by a previous hop has local_id >= 0.
-    flags[i] =
-        (t.local_id[slot] < 0 && t.first_idx[slot] == (uint64_t)i) ? 1 : 0;
+    flags[i] = (t.local_id[slot] < 0 &&
+                t.first_idx[slot] == (uint64_t)(idx_base + i))
+                   ? 1
+                   : 0;
   }
 }
 
@@ -179,6 +186,58 @@ class HIPInducer {
     return insert_and_assign(ids);
   }
 
+  // Staged insert: begin() launches insert+flag+scan with NO host sync
+  // (idx_base keeps first-occurrence order across back-to-back begins);
+  // the caller batches the n_new reads, then commit() assigns ids.
+  // Must be called once per staged hop with the hop's total incoming
+  // id count BEFORE any insert_begin: a table growth between begins
+  // would rehash away the pending (not-yet-assigned) keys.
+  void reserve_incoming(int64_t total) {
+    TORCH_CHECK(pending_ == 0, "reserve_incoming during a staged hop");
+    ensure_capacity(count_ + total, false);
+  }
+
+  std::tuple<torch::Tensor, torch::Tensor> insert_begin(
+      const torch::Tensor& ids, int64_t idx_base) {
+    const int64_t n = ids.size(0);
+    TORCH_CHECK((count_ + pending_ + n) * 2 <= capacity_,
+                "insert_begin: call reserve_incoming for the hop first");
+    auto flags = torch::empty({n}, ids.options());
+    if (n > 0) {
+      auto stream = current_stream();
+      hipLaunchKernelGGL(insert_kernel, dim3(grid_for(n)), dim3(kBlock), 0,
+                         stream, view(), ids.data_ptr<int64_t>(), n,
+                         idx_base);
+      hipLaunchKernelGGL(flag_first_kernel, dim3(grid_for(n)),
+                         dim3(kBlock), 0, stream, view(),
+                         ids.data_ptr<int64_t>(), n, idx_base,
+                         flags.data_ptr<int64_t>());
+    }
+    auto ranks = torch::cumsum(flags, 0);
+    pending_ += n;  // growth headroom until the commits land
+    return {flags, ranks};
+  }
+
+  torch::Tensor insert_commit(const torch::Tensor& ids,
+                              const torch::Tensor& flags,
+                              const torch::Tensor& ranks, int64_t n_new) {
+    const int64_t n = ids.size(0);
+    auto uniq = torch::empty({n_new}, torch::TensorOptions()
+                                          .dtype(torch::kInt64)
+                                          .device(device_));
+    if (n > 0 && n_new > 0) {
+      hipLaunchKernelGGL(assign_kernel, dim3(grid_for(n)), dim3(kBlock), 0,
+                         current_stream(), view(), ids.data_ptr<int64_t>(),
+                         n, ranks.data_ptr<int64_t>(),
+                         flags.data_ptr<int64_t>(), count_,
+                         uniq.data_ptr<int64_t>());
+    }
+    count_ += n_new;
+    pending_ = std::max<int64_t>(0, pending_ - n);
+    if (n_new > 0) node_chunks_.push_back(uniq);
+    return uniq;
+  }
+
   // Relabel arbitrary global ids through the current table (-1 if absent).
   torch::Tensor lookup(const torch::Tensor& ids) {
     auto out = torch::empty_like(ids);
@@ -233,11 +292,12 @@ class HIPInducer {
     if (n == 0) return;
     auto stream = current_stream();
     hipLaunchKernelGGL(insert_kernel, dim3(grid_for(n)), dim3(kBlock), 0,
-                       stream, view(), nodes.data_ptr<int64_t>(), n);
+                       stream, view(), nodes.data_ptr<int64_t>(), n,
+                       (int64_t)0);
     auto flags = torch::empty({n}, nodes.options());
     hipLaunchKernelGGL(flag_first_kernel, dim3(grid_for(n)), dim3(kBlock), 0,
                        stream, view(), nodes.data_ptr<int64_t>(), n,
-                       flags.data_ptr<int64_t>());
+                       (int64_t)0, flags.data_ptr<int64_t>());
     auto ranks = torch::cumsum(flags, 0);
     auto uniq = torch::empty({n}, nodes.options());
     hipLaunchKernelGGL(assign_kernel, dim3(grid_for(n)), dim3(kBlock), 0,
@@ -254,11 +314,12 @@ class HIPInducer {
                                    .device(device_));
     auto stream = current_stream();
     hipLaunchKernelGGL(insert_kernel, dim3(grid_for(n)), dim3(kBlock), 0,
-                       stream, view(), ids.data_ptr<int64_t>(), n);
+                       stream, view(), ids.data_ptr<int64_t>(), n,
+                       (int64_t)0);
     auto flags = torch::empty({n}, ids.options());
     hipLaunchKernelGGL(flag_first_kernel, dim3(grid_for(n)), dim3(kBlock), 0,
                        stream, view(), ids.data_ptr<int64_t>(), n,
-                       flags.data_ptr<int64_t>());
+                       (int64_t)0, flags.data_ptr<int64_t>());
     auto ranks = torch::cumsum(flags, 0);
     const int64_t n_new = ranks[n - 1].item<int64_t>();  // sync
     auto uniq = torch::empty({n_new}, ids.options());
@@ -293,6 +354,7 @@ class HIPInducer {
   std::vector<torch::Tensor> node_chunks_;
   int64_t capacity_ = 0;
   int64_t count_ = 0;
+  int64_t pending_ = 0;  // begun-but-uncommitted inserts (growth guard)
   int64_t reserve_;
 };
 
@@ -316,6 +378,20 @@ torch::Tensor hip_inducer_insert(HIPInducer* ind, const torch::Tensor& ids) {
   return ind->insert(ids);
 }
 int64_t hip_inducer_count(HIPInducer* ind) { return ind->count(); }
+void hip_inducer_reserve(HIPInducer* ind, int64_t total) {
+  ind->reserve_incoming(total);
+}
+std::tuple<torch::Tensor, torch::Tensor> hip_inducer_insert_begin(
+    HIPInducer* ind, const torch::Tensor& ids, int64_t idx_base) {
+  return ind->insert_begin(ids, idx_base);
+}
+torch::Tensor hip_inducer_insert_commit(HIPInducer* ind,
+                                        const torch::Tensor& ids,
+                                        const torch::Tensor& flags,
+                                        const torch::Tensor& ranks,
+                                        int64_t n_new) {
+  return ind->insert_commit(ids, flags, ranks, n_new);
+}
 
 // ---------------------------------------------------------------------------
 // Node subgraph (full induced edge set among a node set).

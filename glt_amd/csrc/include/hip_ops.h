@@ -57,6 +57,15 @@ hip_inducer_induce_next(HIPInducer* ind, const torch::Tensor& srcs,
                         const torch::Tensor& nbrs_num);
 torch::Tensor hip_inducer_lookup(HIPInducer* ind, const torch::Tensor& ids);
 torch::Tensor hip_inducer_insert(HIPInducer* ind, const torch::Tensor& ids);
+// Staged inserts (one totals sync per hop; see HIPInducer comments).
+void hip_inducer_reserve(HIPInducer* ind, int64_t total);
+std::tuple<torch::Tensor, torch::Tensor> hip_inducer_insert_begin(
+    HIPInducer* ind, const torch::Tensor& ids, int64_t idx_base);
+torch::Tensor hip_inducer_insert_commit(HIPInducer* ind,
+                                        const torch::Tensor& ids,
+                                        const torch::Tensor& flags,
+                                        const torch::Tensor& ranks,
+                                        int64_t n_new);
 int64_t hip_inducer_count(HIPInducer* ind);
 
 // --- deferred-sync multi-hop sampler (hip_deferred.hip) ---------------------

@@ -66,6 +66,46 @@ class _PyHeteroInducer:
                              if ntype in self._uniq else fresh)
         return fresh
 
+    def insert_staged(self, batch):
+        """Insert [(ntype, ids), ...] of ONE hop with a single host sync
+        (per-type index bias keeps the fresh-node order identical to
+        sequential inserts).  Falls back to sequential insert for CPU
+        inducers and first-touch types."""
+        if not batch:
+            return []
+        stageable = []
+        for ntype, ids in batch:
+            ind = self._get(ntype)
+            stageable.append(ntype in self._uniq
+                             and hasattr(ind, "insert_begin")
+                             and ids.is_cuda)
+        if not all(stageable):
+            return [self.insert(nt, ids) for nt, ids in batch]
+        totals: Dict[NodeType, int] = {}
+        for ntype, ids in batch:
+            totals[ntype] = totals.get(ntype, 0) + ids.numel()
+        for ntype, tot in totals.items():
+            self._inducers[ntype].reserve_incoming(tot)
+        base: Dict[NodeType, int] = {}
+        pend = []
+        for ntype, ids in batch:
+            b = base.get(ntype, 0)
+            flags, ranks = self._inducers[ntype].insert_begin(ids, b)
+            base[ntype] = b + ids.numel()
+            pend.append((ntype, ids, flags, ranks))
+        dev = pend[0][1].device
+        zero = torch.zeros(1, dtype=torch.long, device=dev)
+        n_new = torch.cat(
+            [(p[3][-1:] if p[3].numel() else zero) for p in pend]
+        ).cpu()  # the hop's single insert sync
+        fresh_list = []
+        for (ntype, ids, flags, ranks), nn in zip(pend, n_new.tolist()):
+            fresh = self._inducers[ntype].insert_commit(ids, flags, ranks,
+                                                       int(nn))
+            self._uniq[ntype] = torch.cat([self._uniq[ntype], fresh])
+            fresh_list.append(fresh)
+        return fresh_list
+
     def lookup(self, ntype, ids: torch.Tensor) -> torch.Tensor:
         ind = self._inducers[ntype]
         if hasattr(ind, "lookup"):
@@ -362,10 +402,14 @@ class NeighborSampler(BaseSampler):
                     continue
                 reqs.append((etype, srcs, k))
             hop_results = self._sample_hop_batched(reqs)
-            # Phase 1: insert all new nodes (deterministic etype order).
-            for etype, srcs, out in hop_results:
-                into = etype[2] if self.edge_dir == "out" else etype[0]
-                fresh = inducer.insert(into, out.nbr)
+            # Phase 1: insert all new nodes (deterministic etype order;
+            # staged — one host sync for the whole hop's inserts).
+            intos = [etype[2] if self.edge_dir == "out" else etype[0]
+                     for etype, _, _ in hop_results]
+            fresh_all = inducer.insert_staged(
+                [(into, out.nbr)
+                 for into, (_, _, out) in zip(intos, hop_results)])
+            for into, fresh in zip(intos, fresh_all):
                 out_nodes.setdefault(into, []).append(fresh)
                 num_nodes.setdefault(into, []).append(fresh.numel())
                 next_frontier.setdefault(into, []).append(fresh)
