@@ -150,6 +150,8 @@ class HIPInducer {
   torch::Tensor init_node(const torch::Tensor& seeds) {
     device_ = seeds.device();
     count_ = 0;
+    pending_ = 0;  // clears a staged hop aborted mid-flight (pooled
+                   // inducers would otherwise refuse reserve_incoming)
     node_chunks_.clear();
     ensure_capacity(std::max<int64_t>(seeds.size(0), reserve_), true);
     return insert_and_assign(seeds);
