@@ -64,7 +64,7 @@ def main():
         ds, fanout, edge_label_index=eli,
         neg_sampling=NegativeSampling("binary"),
         batch_size=args.batch_size, shuffle=True, device=device,
-        to_device=device)
+        to_device=device, prefetch=3 if torch.cuda.is_available() else 0)
     model = GraphSAGE(args.feat_dim, args.hidden, len(fanout)).to(device)
     opt = torch.optim.Adam(model.parameters(), lr=1e-3)
     it = iter(loader)

@@ -48,7 +48,8 @@ class LinkLoader(NodeLoader):
                  neg_sampling: Optional[Union[NegativeSampling, str]] = None,
                  batch_size: int = 1, shuffle: bool = False,
                  drop_last: bool = False, with_edge: bool = False,
-                 to_device: Optional[torch.device] = None):
+                 to_device: Optional[torch.device] = None,
+                 prefetch: int = 0):
         self.data = data
         self.sampler = link_sampler
         if isinstance(edge_label_index, tuple):
@@ -66,24 +67,49 @@ class LinkLoader(NodeLoader):
         self.drop_last = drop_last
         self.with_edge = with_edge
         self.to_device = to_device
+        self.prefetch = prefetch
+        self._prefetcher = None
 
     def __iter__(self):
         self._it = _EdgeSeedIterator(self.edge_row, self.edge_col,
                                      self.edge_label, self.batch_size,
                                      self.shuffle, self.drop_last)
+        if self._prefetcher is not None:
+            self._prefetcher.stop()
+            self._prefetcher = None
+        if self.prefetch > 0 and torch.cuda.is_available():
+            from .node_loader import _Prefetcher
+
+            self._prefetcher = _Prefetcher(self, self._it, self.prefetch)
         return self
 
     def __len__(self):
         return len(_EdgeSeedIterator(self.edge_row, self.edge_col, None,
                                      self.batch_size, False, self.drop_last))
 
-    def __next__(self):
-        row, col, label = next(self._it)
+    def _produce(self, item):
+        row, col, label = item
         inp = EdgeSamplerInput(row=row, col=col, label=label,
                                input_type=self.input_type,
                                neg_sampling=self.neg_sampling)
         out = self.sampler.sample_from_edges(inp)
         return self._collate_fn(out)
+
+    def __next__(self):
+        if self._prefetcher is not None:
+            return self._prefetcher.next()
+        return self._produce(next(self._it))
+
+    def shutdown(self):
+        if self._prefetcher is not None:
+            self._prefetcher.stop()
+            self._prefetcher = None
+
+    def __del__(self):
+        try:
+            self.shutdown()
+        except Exception:
+            pass
 
 
 class LinkNeighborLoader(LinkLoader):
@@ -95,7 +121,7 @@ class LinkNeighborLoader(LinkLoader):
                  device: Optional[torch.device] = None,
                  to_device: Optional[torch.device] = None,
                  edge_dir: Optional[str] = None, seed: Optional[int] = None,
-                 **kwargs):
+                 prefetch: int = 0, **kwargs):
         edge_dir = edge_dir or data.edge_dir
         sampler = NeighborSampler(
             data.get_graph() if not isinstance(data.graph, dict)
@@ -104,4 +130,4 @@ class LinkNeighborLoader(LinkLoader):
             with_weight=with_weight, edge_dir=edge_dir, seed=seed)
         super().__init__(data, sampler, edge_label_index, edge_label,
                          neg_sampling, batch_size, shuffle, drop_last,
-                         with_edge, to_device)
+                         with_edge, to_device, prefetch=prefetch)
