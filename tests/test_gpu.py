@@ -1048,3 +1048,33 @@ def test_staged_sampling_matches_classic():
                                           offs, total)
     assert torch.equal(num1, counts)
     assert torch.equal(nbrs1, nbrs2)
+
+
+def test_hetero_multi_gat_eight_relations_and_no_bias():
+    """Multi-relation fused layer at the kMaxRel=8 limit, with biasless
+    convs mixed in; 9 relations must fall back (and still be correct —
+    covered by the CPU fallback tests)."""
+    from glt_amd.models.hetero import HeteroConv
+    from glt_amd.models.layers import GATConv
+
+    torch.manual_seed(6)
+    n = {"a": 40, "b": 30}
+    x = {t: torch.randn(n[t], 16, device="cuda") for t in n}
+    for n_rel in (8, 9):
+        ets = [("a", f"r{i}", "b") for i in range(n_rel)]
+        convs = {et: GATConv(16, 8, heads=2,
+                             bias=(i % 2 == 0)).cuda()
+                 for i, et in enumerate(ets)}
+        layer = HeteroConv(convs).cuda()
+        ei = {}
+        for et in ets:
+            E = 3 * n["a"]
+            tgt = torch.sort(torch.randint(0, n["a"], (E,),
+                                           device="cuda")).values
+            src = torch.randint(0, n["b"], (E,), device="cuda")
+            ei[et] = torch.stack([tgt, src])
+        layer.use_multi = True
+        out_m = layer(x, ei)
+        layer.use_multi = False
+        out_s = layer(x, ei)
+        assert (out_m["a"] - out_s["a"]).abs().max() < 1e-4, n_rel
