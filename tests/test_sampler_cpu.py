@@ -252,3 +252,30 @@ def test_weighted_no_replace_cpu():
     heavy_rate = sum((vals == i).any(1).float().mean().item()
                      for i in (1, 2, 3)) / 3
     assert heavy_rate > 0.95, heavy_rate
+
+
+def test_hetero_hop_batched_cpu_fallback_deterministic():
+    """_sample_hop_batched falls back to sequential classic sampling on
+    CPU with identical results under the same seed."""
+    import glt_amd
+    from glt_amd.data import Graph, Topology
+    from glt_amd.sampler import NeighborSampler
+
+    et1 = ("u", "a", "v")
+    et2 = ("u", "b", "w")
+    g = {}
+    for et, n2 in ((et1, 30), (et2, 20)):
+        rows = torch.arange(40).repeat_interleave(3)
+        cols = torch.randint(0, n2, (120,))
+        g[et] = Graph(Topology(torch.stack([rows, cols]), num_nodes=40),
+                      mode="CPU")
+    s = NeighborSampler(g, [2], device=torch.device("cpu"))
+    srcs = torch.arange(10)
+    glt_amd.seed_everything(3)
+    batched = s._sample_hop_batched([(et1, srcs, 2), (et2, srcs, 2)])
+    glt_amd.seed_everything(3)
+    o1 = s.sample_one_hop(srcs, 2, etype=et1)
+    o2 = s.sample_one_hop(srcs, 2, etype=et2)
+    assert torch.equal(batched[0][2].nbr, o1.nbr)
+    assert torch.equal(batched[1][2].nbr, o2.nbr)
+    assert torch.equal(batched[0][2].nbr_num, o1.nbr_num)
